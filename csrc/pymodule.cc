@@ -1,9 +1,446 @@
-// Python bindings for moolib_amd._core (placeholder; full bindings follow).
+// Python bindings for moolib_amd._core.
+//
+// API parity with the reference module `moolib._C` (src/moolib.cc): Rpc,
+// Future, Queue, Broker, Group, AllReduce, RpcDeferredReturn, RpcError,
+// create_uid, set_log_level, set_logging, set_max_threads. Batching
+// variants of define() are layered in Python (moolib_amd/__init__.py).
 #include <torch/extension.h>
 
+#include <atomic>
+
+#include "pybits.h"
 #include "rpc.h"
+#include "serde.h"
 #include "services.h"
 
-PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.doc() = "moolib_amd core runtime";
+namespace mrl {
+
+namespace {
+
+py::object g_rpcErrorType;
+
+// Registry of live Rpc instances for atexit cleanup (reference keeps a leaked
+// Rpc list and cleans it up at exit, moolib.cc:127-183).
+std::mutex g_rpcsMu;
+std::vector<std::weak_ptr<Rpc>> g_rpcs;
+
+void registerRpc(const RpcPtr& r) {
+  std::lock_guard<std::mutex> lk(g_rpcsMu);
+  g_rpcs.push_back(r);
 }
+
+void shutdownAll() {
+  std::vector<RpcPtr> live;
+  {
+    std::lock_guard<std::mutex> lk(g_rpcsMu);
+    for (auto& w : g_rpcs) {
+      if (auto p = w.lock()) live.push_back(p);
+    }
+    g_rpcs.clear();
+  }
+  for (auto& p : live) p->shutdown();
+  globalScheduler().drain();
+}
+
+}  // namespace
+
+py::object rpcErrorType() { return g_rpcErrorType; }
+
+// ------------------------------------------------------------ deferred
+
+// Callable handed to deferred handlers; calling it sends the response.
+class RpcDeferredReturn {
+ public:
+  RpcDeferredReturn() = default;
+  explicit RpcDeferredReturn(RespondFn respond) : respond_(std::move(respond)) {}
+  RpcDeferredReturn(const RpcDeferredReturn&) = delete;
+  RpcDeferredReturn& operator=(const RpcDeferredReturn&) = delete;
+  RpcDeferredReturn(RpcDeferredReturn&& o) noexcept
+      : respond_(std::move(o.respond_)), called_(o.called_) {
+    o.respond_ = nullptr;
+    o.called_ = true;
+  }
+  ~RpcDeferredReturn() {
+    if (respond_ && !called_) {
+      auto r = std::move(respond_);
+      r("deferred return dropped without a response", {}, true);
+    }
+  }
+  void call(py::object value) {
+    if (!respond_ || called_) throw RpcError("deferred return already used");
+    called_ = true;
+    std::vector<at::Tensor> tensors;
+    std::string payload = serializeObject(value, tensors);
+    auto r = std::move(respond_);
+    py::gil_scoped_release rel;
+    r(std::move(payload), std::move(tensors), false);
+  }
+  void error(const std::string& msg) {
+    if (!respond_ || called_) throw RpcError("deferred return already used");
+    called_ = true;
+    auto r = std::move(respond_);
+    py::gil_scoped_release rel;
+    r(msg, {}, true);
+  }
+
+ private:
+  RespondFn respond_;
+  bool called_ = false;
+};
+
+// ----------------------------------------------------------------- Rpc
+
+class RpcWrapper {
+ public:
+  RpcWrapper() {
+    rpc_ = Rpc::create();
+    registerRpc(rpc_);
+  }
+  explicit RpcWrapper(RpcPtr rpc) : rpc_(std::move(rpc)) {}
+  RpcPtr rpc() const { return rpc_; }
+
+  void setName(const std::string& n) { rpc_->setName(n); }
+  std::string getName() { return rpc_->getName(); }
+  void setTimeout(double t) { rpc_->setTimeout(t); }
+  std::vector<std::string> listen(const std::string& addr) {
+    py::gil_scoped_release rel;
+    return rpc_->listen(addr);
+  }
+  void connect(const std::string& addr) {
+    py::gil_scoped_release rel;
+    rpc_->connect(addr);
+  }
+  std::string debugInfo() {
+    py::gil_scoped_release rel;
+    return rpc_->debugInfo();
+  }
+  std::vector<std::string> localAddrs() {
+    py::gil_scoped_release rel;
+    return rpc_->localAddrs();
+  }
+  void setTransports(py::object transports) {
+    // We always run both transports (abstract unix + tcp). Accept the
+    // reference's strings for compatibility; reject unknown ones.
+    for (auto t : transports) {
+      std::string s = py::cast<std::string>(t);
+      if (s != "tcp/ip" && s != "shared memory" && s != "uv" && s != "infiniband" && s != "tcp" &&
+          s != "unix") {
+        throw RpcError("unknown transport: " + s);
+      }
+    }
+  }
+
+  void define(const std::string& name, py::function fn) {
+    auto g = std::make_shared<PyGuard>(fn);
+    rpc_->define(name, [g](Frame f, const std::string& from, RespondFn respond) {
+      std::string payload;
+      std::vector<at::Tensor> tensors;
+      bool isErr = false;
+      {
+        py::gil_scoped_acquire gil;
+        try {
+          auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
+          py::object r = g->obj(*args, **kwargs);
+          payload = serializeObject(r, tensors);
+        } catch (py::error_already_set& e) {
+          payload = e.what();
+          isErr = true;
+        } catch (const std::exception& e) {
+          payload = e.what();
+          isErr = true;
+        }
+      }
+      respond(std::move(payload), std::move(tensors), isErr);
+    });
+  }
+
+  void defineDeferred(const std::string& name, py::function fn) {
+    auto g = std::make_shared<PyGuard>(fn);
+    rpc_->define(name, [g](Frame f, const std::string& from, RespondFn respond) {
+      py::gil_scoped_acquire gil;
+      try {
+        auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
+        py::object deferred = py::cast(RpcDeferredReturn(std::move(respond)));
+        g->obj(deferred, *args, **kwargs);
+      } catch (py::error_already_set& e) {
+        // The deferred return (if not moved from) responds with an error when
+        // it is destroyed; report the handler failure too.
+        MRL_LOG_ERROR("deferred handler raised: %s", e.what());
+      } catch (const std::exception& e) {
+        MRL_LOG_ERROR("deferred handler raised: %s", e.what());
+      }
+    });
+  }
+
+  PyQueue defineQueue(const std::string& name) {
+    PyQueue q;
+    rpc_->define(name, [q](Frame f, const std::string& from, RespondFn respond) mutable {
+      py::gil_scoped_acquire gil;
+      try {
+        auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
+        py::object deferred = py::cast(RpcDeferredReturn(std::move(respond)));
+        q.enqueue(py::make_tuple(deferred, args, kwargs));
+      } catch (py::error_already_set& e) {
+        MRL_LOG_ERROR("queue handler failed: %s", e.what());
+      } catch (const std::exception& e) {
+        MRL_LOG_ERROR("queue handler failed: %s", e.what());
+      }
+    });
+    return q;
+  }
+
+  void undefine(const std::string& name) { rpc_->undefine(name); }
+
+  PyFuture asyncCall(const std::string& peer, const std::string& func, py::args args,
+                     py::kwargs kwargs) {
+    std::vector<at::Tensor> tensors;
+    std::string payload = serializeCall(args, kwargs, tensors);
+    PyFuture fut;
+    auto st = fut.state();
+    py::gil_scoped_release rel;
+    rpc_->sendRequest(peer, func, std::move(payload), std::move(tensors),
+                      [st](Frame* resp, const std::string* err) {
+                        py::gil_scoped_acquire gil;
+                        if (err) {
+                          futureFail(st, *err);
+                        } else {
+                          try {
+                            py::object v = deserializeObject(resp->payload, resp->tensors);
+                            futureComplete(st, std::move(v));
+                          } catch (py::error_already_set& e) {
+                            futureFail(st, std::string("deserialize failed: ") + e.what());
+                          } catch (const std::exception& e) {
+                            futureFail(st, std::string("deserialize failed: ") + e.what());
+                          }
+                        }
+                      });
+    return fut;
+  }
+
+  void asyncCallback(const std::string& peer, const std::string& func, py::function cb,
+                     py::args args, py::kwargs kwargs) {
+    std::vector<at::Tensor> tensors;
+    std::string payload = serializeCall(args, kwargs, tensors);
+    auto g = std::make_shared<PyGuard>(cb);
+    py::gil_scoped_release rel;
+    rpc_->sendRequest(peer, func, std::move(payload), std::move(tensors),
+                      [g](Frame* resp, const std::string* err) {
+                        py::gil_scoped_acquire gil;
+                        try {
+                          if (err) {
+                            g->obj(py::none(), rpcErrorType()(*err));
+                          } else {
+                            py::object v = deserializeObject(resp->payload, resp->tensors);
+                            g->obj(v, py::none());
+                          }
+                        } catch (py::error_already_set& e) {
+                          e.discard_as_unraisable("moolib_amd async_callback");
+                        }
+                      });
+  }
+
+  py::object syncCall(const std::string& peer, const std::string& func, py::args args,
+                      py::kwargs kwargs) {
+    PyFuture fut = asyncCall(peer, func, args, kwargs);
+    return fut.resultNoTimeout();
+  }
+
+ private:
+  RpcPtr rpc_;
+};
+
+// -------------------------------------------------------------- Broker
+
+class BrokerWrapper {
+ public:
+  explicit BrokerWrapper(py::object rpc) {
+    if (rpc.is_none()) {
+      rpc_ = Rpc::create();
+      rpc_->setName("broker");
+      registerRpc(rpc_);
+    } else {
+      rpc_ = py::cast<RpcWrapper&>(rpc).rpc();
+    }
+    broker_ = std::make_unique<Broker>(rpc_);
+  }
+  void setName(const std::string& n) { rpc_->setName(n); }
+  void listen(const std::string& addr) {
+    py::gil_scoped_release rel;
+    rpc_->listen(addr);
+  }
+  void update() { broker_->update(); }
+  RpcPtr rpc() { return rpc_; }
+
+ private:
+  RpcPtr rpc_;
+  std::unique_ptr<Broker> broker_;
+};
+
+// --------------------------------------------------------------- Group
+
+PyFold makePyFold(PyGuardPtr op) {
+  return [op](ReduceValue& dst, ReduceValue& src) {
+    py::gil_scoped_acquire gil;
+    py::object a = deserializeObject(dst.payload, dst.tensors);
+    py::object b = deserializeObject(src.payload, src.tensors);
+    py::object r;
+    if (op && op->obj.ptr() && !op->obj.is_none()) {
+      r = op->obj(a, b);
+      if (r.is_none()) r = a;  // ops that mutate dst in place and return None
+    } else {
+      PyObject* res = PyNumber_Add(a.ptr(), b.ptr());
+      if (!res) throw py::error_already_set();
+      r = py::reinterpret_steal<py::object>(res);
+    }
+    dst.tensors.clear();
+    dst.payload = serializeObject(r, dst.tensors);
+  };
+}
+
+class GroupWrapper {
+ public:
+  GroupWrapper(RpcWrapper& rpc, const std::string& name) {
+    group_ = Group::create(rpc.rpc(), name);
+  }
+  std::shared_ptr<Group> group() { return group_; }
+
+  void update() { group_->update(); }
+  std::vector<std::string> members() { return group_->members(); }
+  uint64_t syncId() { return group_->syncId(); }
+  std::string name() { return group_->name(); }
+  bool active() { return group_->active(); }
+  void setBrokerName(const std::string& n) { group_->setBrokerName(n); }
+  void setTimeout(double t) { group_->setTimeout(t); }
+  void setSortOrder(int64_t o) { group_->setSortOrder(o); }
+
+  PyFuture allReduce(const std::string& name, py::object value, py::object op) {
+    ReduceValue v;
+    v.kind = ReduceValue::pyObject;
+    v.payload = serializeObject(value, v.tensors);
+    auto opGuard = std::make_shared<PyGuard>(op);
+    PyFuture fut;
+    auto st = fut.state();
+    PyFold fold = makePyFold(opGuard);
+    py::gil_scoped_release rel;
+    group_->allReduce(name, std::move(v), std::move(fold),
+                      [st](ReduceValue* rv, const std::string* err) {
+                        py::gil_scoped_acquire gil;
+                        if (err) {
+                          futureFail(st, *err);
+                        } else {
+                          try {
+                            futureComplete(st, deserializeObject(rv->payload, rv->tensors));
+                          } catch (py::error_already_set& e) {
+                            futureFail(st, std::string("deserialize failed: ") + e.what());
+                          } catch (const std::exception& e) {
+                            futureFail(st, e.what());
+                          }
+                        }
+                      });
+    return fut;
+  }
+
+ private:
+  std::shared_ptr<Group> group_;
+};
+
+// -------------------------------------------------------------- module
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "moolib_amd core runtime (MI355X-native distributed RL framework)";
+
+  auto rpcError = py::register_exception<RpcError>(m, "RpcError", PyExc_RuntimeError);
+  g_rpcErrorType = rpcError;
+
+  m.def("create_uid", [] { return randomUid(); });
+  m.def("set_log_level", [](const std::string& level) {
+    int v = level == "none"      ? 0
+            : level == "error"   ? 1
+            : level == "info"    ? 2
+            : level == "verbose" ? 3
+            : level == "debug"   ? 4
+                                 : -1;
+    if (v < 0) throw RpcError("unknown log level: " + level);
+    g_logLevel.store(v);
+  });
+  m.def("set_logging", [](py::object sink) {
+    if (sink.is_none()) {
+      g_logSink = nullptr;
+      return;
+    }
+    auto g = std::make_shared<PyGuard>(sink);
+    g_logSink = [g](int level, const std::string& msg) {
+      if (!Py_IsInitialized()) return;
+      py::gil_scoped_acquire gil;
+      try {
+        g->obj(level, msg);
+      } catch (py::error_already_set& e) {
+        e.discard_as_unraisable("moolib_amd log sink");
+      }
+    };
+  });
+  m.def("set_max_threads", [](int n) { globalScheduler().setMaxThreads(n); });
+  m.def("_shutdown_all", [] {
+    py::gil_scoped_release rel;
+    shutdownAll();
+  });
+
+  py::class_<PyFuture>(m, "Future")
+      .def("result", &PyFuture::resultNoTimeout)
+      .def("result", &PyFuture::resultTimeout, py::arg("timeout"))
+      .def("wait", &PyFuture::waitNoTimeout)
+      .def("wait", &PyFuture::waitTimeout, py::arg("timeout"))
+      .def("done", &PyFuture::done)
+      .def("cancel", &PyFuture::cancel)
+      .def("exception", &PyFuture::exception)
+      .def("_add_done_callback", &PyFuture::addDoneCallback);
+
+  py::class_<PyQueue>(m, "Queue")
+      .def(py::init<>())
+      .def("enqueue", &PyQueue::enqueue)
+      .def("size", &PyQueue::size)
+      .def("_pop_future", &PyQueue::popFuture);
+
+  py::class_<RpcDeferredReturn>(m, "RpcDeferredReturn")
+      .def("__call__", &RpcDeferredReturn::call)
+      .def("error", &RpcDeferredReturn::error);
+
+  py::class_<RpcWrapper>(m, "Rpc")
+      .def(py::init<>())
+      .def("set_name", &RpcWrapper::setName, py::arg("name"))
+      .def("get_name", &RpcWrapper::getName)
+      .def("set_timeout", &RpcWrapper::setTimeout, py::arg("timeout"))
+      .def("set_transports", &RpcWrapper::setTransports, py::arg("transports"))
+      .def("listen", &RpcWrapper::listen, py::arg("address"))
+      .def("connect", &RpcWrapper::connect, py::arg("address"))
+      .def("local_addrs", &RpcWrapper::localAddrs)
+      .def("debug_info", &RpcWrapper::debugInfo)
+      .def("_define_raw", &RpcWrapper::define)
+      .def("define_deferred_raw", &RpcWrapper::defineDeferred)
+      .def("define_queue_raw", &RpcWrapper::defineQueue)
+      .def("undefine", &RpcWrapper::undefine, py::arg("name"))
+      .def("async_", &RpcWrapper::asyncCall)
+      .def("async_callback", &RpcWrapper::asyncCallback)
+      .def("sync", &RpcWrapper::syncCall);
+
+  py::class_<BrokerWrapper>(m, "Broker")
+      .def(py::init<py::object>(), py::arg("rpc") = py::none())
+      .def("set_name", &BrokerWrapper::setName, py::arg("name"))
+      .def("listen", &BrokerWrapper::listen, py::arg("address"))
+      .def("update", &BrokerWrapper::update, py::call_guard<py::gil_scoped_release>());
+
+  py::class_<GroupWrapper>(m, "Group")
+      .def(py::init<RpcWrapper&, const std::string&>(), py::arg("rpc"), py::arg("name"))
+      .def("update", &GroupWrapper::update, py::call_guard<py::gil_scoped_release>())
+      .def("members", &GroupWrapper::members, py::call_guard<py::gil_scoped_release>())
+      .def("sync_id", &GroupWrapper::syncId, py::call_guard<py::gil_scoped_release>())
+      .def("name", &GroupWrapper::name)
+      .def("active", &GroupWrapper::active, py::call_guard<py::gil_scoped_release>())
+      .def("set_broker_name", &GroupWrapper::setBrokerName)
+      .def("set_timeout", &GroupWrapper::setTimeout)
+      .def("set_sort_order", &GroupWrapper::setSortOrder)
+      .def("all_reduce", &GroupWrapper::allReduce, py::arg("name"), py::arg("value"),
+           py::arg("op") = py::none());
+}
+
+}  // namespace mrl

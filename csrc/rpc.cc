@@ -489,17 +489,24 @@ void Rpc::handleRequest(ConnId id, Frame&& f) {
 
 void Rpc::handleResponse(ConnId id, Frame&& f, bool isError) {
   ResponseCallback cb;
+  std::string peerName, funcName;
   {
     std::lock_guard<std::mutex> lk(mu_);
     auto it = outgoing_.find(f.rid);
     if (it == outgoing_.end()) return;  // duplicate/late response
     cb = std::move(it->second.cb);
+    peerName = it->second.peerName;
+    funcName = it->second.funcName;
     outgoing_.erase(it);
   }
   if (!cb) return;
-  globalScheduler().run([cb = std::move(cb), f = std::move(f), isError]() mutable {
+  globalScheduler().run([cb = std::move(cb), f = std::move(f), isError, peerName, funcName]() mutable {
     if (isError) {
       std::string err = f.payload.empty() ? "remote error" : f.payload;
+      if (err.rfind("unknown function id", 0) == 0) {
+        // Mirror the reference's error text for a missing remote function.
+        err = "RPC remote function " + peerName + "::'" + funcName + "' does not exist";
+      }
       cb(nullptr, &err);
     } else {
       cb(&f, nullptr);
@@ -578,8 +585,8 @@ void Rpc::timerLoop() {
       // Outgoing: timeouts + resend-on-reconnect + discovery retries.
       for (auto& [rid, rec] : outgoing_) {
         if (t >= rec.deadline) {
-          failures.push_back({rid, "timed out waiting for response from '" + rec.peerName + "' (" +
-                                       rec.funcName + ")"});
+          // Message format mirrors the reference's timeout error text.
+          failures.push_back({rid, "Call (" + rec.peerName + "::" + rec.funcName + ") timed out"});
           continue;
         }
         PeerInfo& p = getPeer(rec.peerName);

@@ -1,0 +1,256 @@
+"""moolib_amd — an MI355X-native distributed RL / RPC framework.
+
+A brand-new implementation of the capabilities of facebookresearch/moolib
+(reference: py/moolib/__init__.py), built for AMD Instinct MI355X: the
+peer-to-peer RPC layer runs on an epoll reactor with zero-copy tensor
+frames; gradient all-reduce runs on RCCL over xGMI when peers share a
+torch.distributed world (see moolib_amd.parallel); device math for the
+IMPALA path is hand-written HIP/CDNA4 kernels (moolib_amd.ops).
+"""
+import asyncio
+import atexit
+import threading
+
+import torch
+
+from . import _core
+from ._core import (
+    Broker,
+    Future,
+    Queue,
+    RpcDeferredReturn,
+    RpcError,
+    create_uid,
+    set_log_level,
+    set_logging,
+    set_max_threads,
+)
+from .utils import nest
+
+# Group / AllReduce --------------------------------------------------------
+
+Group = _core.Group
+AllReduce = Future  # group.all_reduce returns a Future-shaped handle
+
+
+# Future / Queue awaitability ---------------------------------------------
+
+
+def _future_await(self):
+    if self.done():
+        exc = self.exception()
+        if exc is not None:
+            raise exc
+        return (yield from _completed(self.result()))
+    loop = asyncio.get_running_loop()
+    af = loop.create_future()
+
+    def _transfer():
+        if af.cancelled():
+            return
+        exc = self.exception()
+        if exc is not None:
+            af.set_exception(exc)
+        else:
+            try:
+                af.set_result(self.result())
+            except Exception as e:  # noqa: BLE001
+                af.set_exception(e)
+
+    def _cb():
+        try:
+            loop.call_soon_threadsafe(_transfer)
+        except RuntimeError:
+            pass  # loop already closed
+
+    self._add_done_callback(_cb)
+    return af.__await__()
+
+
+def _completed(value):
+    return value
+    yield  # pragma: no cover — makes this a generator
+
+
+Future.__await__ = _future_await
+Future.__iter__ = _future_await
+
+
+def _queue_await(self):
+    return self._pop_future().__await__()
+
+
+def _queue_get(self):
+    """Blocking pop (convenience; not in the reference API)."""
+    return self._pop_future().result()
+
+
+Queue.__await__ = _queue_await
+Queue.__iter__ = _queue_await
+Queue.get = _queue_get
+
+
+# Rpc with batched defines --------------------------------------------------
+
+
+class _BatchCollector:
+    """Collects deferred calls until batch_size is reached, then fires.
+
+    Implements the reference's define(batch_size=...) semantics
+    (src/moolib.cc:1007-1178 + batch_utils): tensor leaves are stacked along
+    a new dim 0 and moved to `device`; non-tensor leaves pass through from
+    the first call; returned tensors are split back per caller.
+    """
+
+    def __init__(self, batch_size, device, process, dynamic=False, max_latency=0.01):
+        self.batch_size = batch_size
+        self.device = device
+        self.process = process  # fn(batched_args, batched_kwargs, respond_all)
+        self.dynamic = dynamic
+        self.max_latency = max_latency
+        self.lock = threading.Lock()
+        self.pending = []
+        self.timer = None
+
+    def add(self, deferred, args, kwargs):
+        fire = None
+        with self.lock:
+            self.pending.append((deferred, args, kwargs))
+            if len(self.pending) >= self.batch_size:
+                fire = self.pending
+                self.pending = []
+                if self.timer is not None:
+                    self.timer.cancel()
+                    self.timer = None
+            elif self.dynamic and self.timer is None:
+                self.timer = threading.Timer(self.max_latency, self._flush)
+                self.timer.daemon = True
+                self.timer.start()
+        if fire:
+            self._fire(fire)
+
+    def _flush(self):
+        with self.lock:
+            fire = self.pending
+            self.pending = []
+            self.timer = None
+        if fire:
+            self._fire(fire)
+
+    def _fire(self, batch):
+        n = len(batch)
+        args0, kwargs0 = batch[0][1], batch[0][2]
+
+        def stack_leaves(leaves):
+            if all(isinstance(x, torch.Tensor) for x in leaves):
+                t = torch.stack(leaves)
+                if self.device is not None:
+                    t = t.to(self.device)
+                return t
+            return leaves[0]
+
+        batched_args = nest.map_many(stack_leaves, *[b[1] for b in batch]) if args0 else ()
+        batched_kwargs = (
+            nest.map_many(stack_leaves, *[b[2] for b in batch]) if kwargs0 else {}
+        )
+
+        def respond_all(result):
+            for i, (deferred, _, _) in enumerate(batch):
+                out_i = nest.map(
+                    lambda t, i=i: t[i].cpu() if isinstance(t, torch.Tensor) else t, result
+                )
+                deferred(out_i)
+
+        self.process(tuple(batched_args), dict(batched_kwargs), respond_all, n)
+
+
+class Rpc(_core.Rpc):
+    """Named RPC peer (reference API: src/moolib.cc Rpc bindings)."""
+
+    def define(self, name, func, batch_size=None, device=None, dynamic_batching=False):
+        if batch_size is None:
+            if device is None:
+                self._define_raw(name, func)
+            else:
+
+                def moved(*args, **kwargs):
+                    args = nest.map(
+                        lambda t: t.to(device) if isinstance(t, torch.Tensor) else t, args
+                    )
+                    kwargs = nest.map(
+                        lambda t: t.to(device) if isinstance(t, torch.Tensor) else t, kwargs
+                    )
+                    return func(*args, **kwargs)
+
+                self._define_raw(name, moved)
+            return
+
+        def process(bargs, bkwargs, respond_all, n):
+            try:
+                result = func(*bargs, **bkwargs)
+            except Exception as e:  # noqa: BLE001
+                raise e
+            respond_all(result)
+
+        collector = _BatchCollector(batch_size, device, process, dynamic=dynamic_batching)
+        self.define_deferred_raw(name, lambda d, *a, **kw: collector.add(d, a, kw))
+
+    def define_deferred(self, name, func, batch_size=None, device=None, dynamic_batching=False):
+        if batch_size is None:
+            self.define_deferred_raw(name, func)
+            return
+
+        def process(bargs, bkwargs, respond_all, n):
+            func(respond_all, *bargs, **bkwargs)
+
+        collector = _BatchCollector(batch_size, device, process, dynamic=dynamic_batching)
+        self.define_deferred_raw(name, lambda d, *a, **kw: collector.add(d, a, kw))
+
+    def define_queue(self, name, batch_size=None, device=None, dynamic_batching=False):
+        if batch_size is None:
+            return self.define_queue_raw(name)
+        queue = Queue()
+
+        def process(bargs, bkwargs, respond_all, n):
+            queue.enqueue((respond_all, bargs, bkwargs))
+
+        collector = _BatchCollector(batch_size, device, process, dynamic=dynamic_batching)
+        self.define_deferred_raw(name, lambda d, *a, **kw: collector.add(d, a, kw))
+        return queue
+
+
+# Training components (bound as the C++ layer grows) -----------------------
+
+try:
+    from ._core import Accumulator, Batcher  # noqa: F401
+except ImportError:  # pragma: no cover — during staged bring-up
+    pass
+
+try:
+    from ._core import EnvPool, EnvRunner, EnvStepper, EnvStepperFuture  # noqa: F401
+except ImportError:  # pragma: no cover
+    pass
+
+
+atexit.register(_core._shutdown_all)
+
+__all__ = [
+    "Accumulator",
+    "AllReduce",
+    "Batcher",
+    "Broker",
+    "EnvPool",
+    "EnvRunner",
+    "EnvStepper",
+    "EnvStepperFuture",
+    "Future",
+    "Group",
+    "Queue",
+    "Rpc",
+    "RpcDeferredReturn",
+    "RpcError",
+    "create_uid",
+    "set_log_level",
+    "set_logging",
+    "set_max_threads",
+]
