@@ -22,13 +22,18 @@ namespace mrl {
 
 namespace py = pybind11;
 
+// True while the interpreter can safely be entered from a non-Python thread.
+// During finalization, acquiring the GIL from a C++ thread aborts the
+// process, so every completion/cleanup path checks this first.
+inline bool pyAlive() { return Py_IsInitialized() && !_Py_IsFinalizing(); }
+
 struct PyGuard {
   py::object obj;
   explicit PyGuard(py::object o) : obj(std::move(o)) {}
   PyGuard(const PyGuard&) = delete;
   ~PyGuard() {
     if (obj.ptr() != nullptr) {
-      if (Py_IsInitialized()) {
+      if (pyAlive()) {
         py::gil_scoped_acquire gil;
         obj = py::object();
       } else {
@@ -54,7 +59,7 @@ struct FutureState {
 
   ~FutureState() {
     if ((value.ptr() || !callbacks.empty())) {
-      if (Py_IsInitialized()) {
+      if (pyAlive()) {
         py::gil_scoped_acquire gil;
         value = py::object();
         callbacks.clear();

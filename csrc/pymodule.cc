@@ -133,6 +133,7 @@ class RpcWrapper {
   void define(const std::string& name, py::function fn) {
     auto g = std::make_shared<PyGuard>(fn);
     rpc_->define(name, [g](Frame f, const std::string& from, RespondFn respond) {
+      if (!pyAlive()) return;
       std::string payload;
       std::vector<at::Tensor> tensors;
       bool isErr = false;
@@ -157,6 +158,7 @@ class RpcWrapper {
   void defineDeferred(const std::string& name, py::function fn) {
     auto g = std::make_shared<PyGuard>(fn);
     rpc_->define(name, [g](Frame f, const std::string& from, RespondFn respond) {
+      if (!pyAlive()) return;
       py::gil_scoped_acquire gil;
       try {
         auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
@@ -175,6 +177,7 @@ class RpcWrapper {
   PyQueue defineQueue(const std::string& name) {
     PyQueue q;
     rpc_->define(name, [q](Frame f, const std::string& from, RespondFn respond) mutable {
+      if (!pyAlive()) return;
       py::gil_scoped_acquire gil;
       try {
         auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
@@ -200,6 +203,7 @@ class RpcWrapper {
     py::gil_scoped_release rel;
     rpc_->sendRequest(peer, func, std::move(payload), std::move(tensors),
                       [st](Frame* resp, const std::string* err) {
+                        if (!pyAlive()) return;
                         py::gil_scoped_acquire gil;
                         if (err) {
                           futureFail(st, *err);
@@ -225,6 +229,7 @@ class RpcWrapper {
     py::gil_scoped_release rel;
     rpc_->sendRequest(peer, func, std::move(payload), std::move(tensors),
                       [g](Frame* resp, const std::string* err) {
+                        if (!pyAlive()) return;
                         py::gil_scoped_acquire gil;
                         try {
                           if (err) {
@@ -280,6 +285,7 @@ class BrokerWrapper {
 
 PyFold makePyFold(PyGuardPtr op) {
   return [op](ReduceValue& dst, ReduceValue& src) {
+    if (!pyAlive()) throw RpcError("interpreter shutting down");
     py::gil_scoped_acquire gil;
     py::object a = deserializeObject(dst.payload, dst.tensors);
     py::object b = deserializeObject(src.payload, src.tensors);
@@ -324,6 +330,7 @@ class GroupWrapper {
     py::gil_scoped_release rel;
     group_->allReduce(name, std::move(v), std::move(fold),
                       [st](ReduceValue* rv, const std::string* err) {
+                        if (!pyAlive()) return;
                         py::gil_scoped_acquire gil;
                         if (err) {
                           futureFail(st, *err);
@@ -370,7 +377,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     }
     auto g = std::make_shared<PyGuard>(sink);
     g_logSink = [g](int level, const std::string& msg) {
-      if (!Py_IsInitialized()) return;
+      if (!pyAlive()) return;
       py::gil_scoped_acquire gil;
       try {
         g->obj(level, msg);

@@ -117,7 +117,10 @@ void Rpc::connect(const std::string& addr) {
   e.addr = addr;
   e.lastAttempt = now();
   e.conn = engine_->connect(addr);
-  conns_[e.conn] = ConnInfo{};
+  ConnInfo ci;
+  ci.established = now();
+  ci.lastRecv = now();
+  conns_[e.conn] = ci;
   endpoints_.push_back(e);
 }
 
@@ -292,6 +295,8 @@ void Rpc::tryConnectPeerLocked(const std::string& name, PeerInfo& p) {
   p.connecting = engine_->connect(addr);
   ConnInfo ci;
   ci.peerName = name;
+  ci.established = now();
+  ci.lastRecv = now();
   conns_[p.connecting] = ci;
 }
 
@@ -579,7 +584,10 @@ void Rpc::timerLoop() {
           e.lastAttempt = t;
           e.backoff = std::min(e.backoff * 1.6, 2.0);
           e.conn = engine_->connect(e.addr);
-          conns_[e.conn] = ConnInfo{};
+          ConnInfo ci;
+          ci.established = now();
+          ci.lastRecv = now();
+          conns_[e.conn] = ci;
         }
       }
       // Outgoing: timeouts + resend-on-reconnect + discovery retries.

@@ -37,39 +37,26 @@ AllReduce = Future  # group.all_reduce returns a Future-shaped handle
 
 
 def _future_await(self):
-    if self.done():
-        exc = self.exception()
-        if exc is not None:
-            raise exc
-        return (yield from _completed(self.result()))
-    loop = asyncio.get_running_loop()
-    af = loop.create_future()
+    if not self.done():
+        loop = asyncio.get_running_loop()
+        af = loop.create_future()
 
-    def _transfer():
-        if af.cancelled():
-            return
-        exc = self.exception()
-        if exc is not None:
-            af.set_exception(exc)
-        else:
+        def _transfer():
+            if not af.cancelled():
+                af.set_result(None)
+
+        def _cb():
             try:
-                af.set_result(self.result())
-            except Exception as e:  # noqa: BLE001
-                af.set_exception(e)
+                loop.call_soon_threadsafe(_transfer)
+            except RuntimeError:
+                pass  # loop already closed
 
-    def _cb():
-        try:
-            loop.call_soon_threadsafe(_transfer)
-        except RuntimeError:
-            pass  # loop already closed
-
-    self._add_done_callback(_cb)
-    return af.__await__()
-
-
-def _completed(value):
-    return value
-    yield  # pragma: no cover — makes this a generator
+        self._add_done_callback(_cb)
+        yield from af.__await__()
+    exc = self.exception()
+    if exc is not None:
+        raise exc
+    return self.result()
 
 
 Future.__await__ = _future_await

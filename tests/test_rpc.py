@@ -1,0 +1,230 @@
+"""RPC core tests (loopback, multi-peer in one process).
+
+Mirrors the coverage of the reference's test/unit/test_simple.py and
+test_tensors.py against our own implementation.
+"""
+import asyncio
+import time
+
+import pytest
+import torch
+
+import moolib_amd
+
+
+def make_pair(timeout=10):
+    host = moolib_amd.Rpc()
+    client = moolib_amd.Rpc()
+    host.set_name("host")
+    client.set_name("client")
+    client.set_timeout(timeout)
+    addr = host.listen("127.0.0.1:0")[0]
+    client.connect(addr)
+    return host, client
+
+
+class TestRpcBasics:
+    def test_async_and_sync(self):
+        host, client = make_pair()
+        calls = 0
+
+        def hello(message):
+            nonlocal calls
+            calls += 1
+            return "response to '%s'" % message
+
+        host.define("hello", hello)
+        fut = client.async_("host", "hello", "msg1")
+        assert fut.result() == "response to 'msg1'"
+        assert client.sync("host", "hello", "msg2") == "response to 'msg2'"
+        assert calls == 2
+
+    def test_kwargs_and_types(self):
+        host, client = make_pair()
+        host.define("echo", lambda *a, **kw: (a, kw))
+        a, kw = client.sync(
+            "host", "echo", 1, 2.5, "s", b"b", None, True, [1, [2]], (3, 4), {"k": 5}, big=2**100
+        )
+        assert a == (1, 2.5, "s", b"b", None, True, [1, [2]], (3, 4), {"k": 5})
+        assert kw == {"big": 2**100}
+
+    def test_tensor_roundtrip(self):
+        host, client = make_pair()
+        host.define("double", lambda t: t * 2)
+        for dtype in (torch.float32, torch.float64, torch.int64, torch.bfloat16, torch.uint8):
+            x = (torch.randn(64, 3) * 10).to(dtype)
+            y = client.sync("host", "double", x)
+            assert y.dtype == dtype
+            assert torch.equal(y, x * 2)
+
+    def test_numpy_roundtrip(self):
+        import numpy as np
+
+        host, client = make_pair()
+        host.define("sum", lambda arr: arr.sum())
+        x = np.arange(12, dtype=np.float32).reshape(3, 4)
+        assert client.sync("host", "sum", x) == pytest.approx(66.0)
+
+    def test_pickle_fallback(self):
+        host, client = make_pair()
+
+        class Thing:
+            def __init__(self, v):
+                self.v = v
+
+            def __eq__(self, o):
+                return self.v == o.v
+
+        # Class must be importable for pickle: use a dict-of-set instead.
+        host.define("echo", lambda x: x)
+        assert client.sync("host", "echo", {1, 2, 3}) == {1, 2, 3}
+
+    def test_async_callback(self):
+        host, client = make_pair()
+        host.define("hello", lambda m: "resp:" + m)
+        results = []
+        client.async_callback("host", "hello", lambda r, e: results.append((r, e)), "x")
+        t0 = time.time()
+        while not results and time.time() - t0 < 10:
+            time.sleep(0.01)
+        assert results == [("resp:x", None)]
+
+    def test_unknown_function(self):
+        host, client = make_pair()
+        host.define("hello", lambda: 42)
+        with pytest.raises(RuntimeError, match="does not exist"):
+            client.sync("host", "missing fn")
+
+    def test_unknown_peer_times_out(self):
+        host, client = make_pair(timeout=1)
+        t0 = time.time()
+        with pytest.raises(RuntimeError, match="timed out"):
+            client.sync("nowhere", "hello")
+        assert time.time() - t0 < 5
+
+    def test_dead_host_resend_on_rebirth(self):
+        host, client = make_pair(timeout=1)
+        host.define("hello", lambda: 42)
+        addr = [a for a in host.local_addrs() if a.startswith("tcp://127")][0]
+        assert client.sync("host", "hello") == 42
+        del host
+        with pytest.raises(RuntimeError, match="timed out"):
+            client.sync("host", "hello")
+        host2 = moolib_amd.Rpc()
+        host2.set_name("host")
+        host2.define("hello", lambda: 43)
+        host2.listen("127.0.0.1:0")
+        client.set_timeout(30)
+        client.connect(host2.local_addrs()[0])
+        assert client.sync("host", "hello") == 43
+
+    def test_exception_in_handler(self):
+        host, client = make_pair()
+
+        def boom():
+            raise ValueError("kaboom")
+
+        host.define("boom", boom)
+        with pytest.raises(RuntimeError, match="kaboom"):
+            client.sync("host", "boom")
+
+    def test_future_exception_api(self):
+        host, client = make_pair(timeout=1)
+        fut = client.async_("nowhere", "fn")
+        fut.wait()
+        assert fut.done()
+        assert isinstance(fut.exception(), moolib_amd.RpcError)
+
+    def test_bidirectional(self):
+        host, client = make_pair()
+        host.define("h", lambda x: x + 1)
+        client.define("c", lambda x: x * 2)
+        assert client.sync("host", "h", 1) == 2
+        assert host.sync("client", "c", 21) == 42
+
+
+class TestDeferredAndQueue:
+    def test_define_deferred(self):
+        host, client = make_pair()
+
+        def handler(callback, msg):
+            callback("deferred:" + msg)
+
+        host.define_deferred("d", handler)
+        assert client.sync("host", "d", "x") == "deferred:x"
+        assert client.sync("host", "d", msg="named") == "deferred:named"
+
+    def test_define_queue(self):
+        host, client = make_pair()
+        queue = host.define_queue("q")
+        fut = client.async_("host", "q", 10, k=3)
+        ret, args, kwargs = queue.get()
+        assert args == (10,)
+        assert kwargs == {"k": 3}
+        ret(args[0] + kwargs["k"])
+        assert fut.result() == 13
+
+    def test_queue_asyncio(self):
+        host, client = make_pair()
+        queue = host.define_queue("q")
+
+        async def serve_and_call():
+            fut = client.async_("host", "q", 5)
+            ret, args, kwargs = await queue
+            ret(args[0] * 3)
+            return await fut
+
+        assert asyncio.run(serve_and_call()) == 15
+
+    def test_future_await(self):
+        host, client = make_pair()
+        host.define("inc", lambda x: x + 1)
+
+        async def call():
+            return await client.async_("host", "inc", 41)
+
+        assert asyncio.run(call()) == 42
+
+
+class TestBatchedDefine:
+    @pytest.mark.parametrize("style", ["define", "deferred", "queue"])
+    def test_batched(self, style):
+        bs = 4
+        host = moolib_amd.Rpc()
+        host.set_name("host")
+        addr = host.listen("127.0.0.1:0")[0]
+
+        def fn(msg, tensor):
+            assert tensor.shape == (bs, 2, 3)
+            return msg, tensor.flatten(1).sum(1)
+
+        if style == "define":
+            host.define("f", fn, batch_size=bs)
+        elif style == "deferred":
+            host.define_deferred("f", lambda cb, m, t: cb(fn(m, t)), batch_size=bs)
+        else:
+            queue = host.define_queue("f", batch_size=bs)
+
+            def pump():
+                ret, args, kwargs = queue.get()
+                ret(fn(*args, **kwargs))
+
+            import threading
+
+            threading.Thread(target=pump, daemon=True).start()
+
+        clients = []
+        futures = []
+        tensors = []
+        for i in range(bs):
+            c = moolib_amd.Rpc()
+            c.set_timeout(15)
+            c.connect(addr)
+            t = torch.randn(2, 3)
+            tensors.append(t)
+            futures.append(c.async_("host", "f", "m", t))
+            clients.append(c)
+        for i, f in enumerate(futures):
+            msg, s = f.result()
+            assert msg == "m"
+            assert torch.allclose(s, tensors[i].sum())
