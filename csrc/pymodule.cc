@@ -17,7 +17,7 @@ namespace mrl {
 
 namespace {
 
-py::object g_rpcErrorType;
+py::object* g_rpcErrorType = new py::object();  // leaked: see serde.cc note
 
 // Registry of live Rpc instances for atexit cleanup (reference keeps a leaked
 // Rpc list and cleans it up at exit, moolib.cc:127-183).
@@ -44,7 +44,7 @@ void shutdownAll() {
 
 }  // namespace
 
-py::object rpcErrorType() { return g_rpcErrorType; }
+py::object rpcErrorType() { return *g_rpcErrorType; }
 
 // ------------------------------------------------------------ deferred
 
@@ -357,7 +357,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "moolib_amd core runtime (MI355X-native distributed RL framework)";
 
   auto rpcError = py::register_exception<RpcError>(m, "RpcError", PyExc_RuntimeError);
-  g_rpcErrorType = rpcError;
+  *g_rpcErrorType = rpcError;
 
   m.def("create_uid", [] { return randomUid(); });
   m.def("set_log_level", [](const std::string& level) {

@@ -23,24 +23,26 @@ enum Tag : uint8_t {
   tDevTensor = 13,  // u32 index + u8 deviceIndex: tensor that lived on an accelerator
 };
 
+// These statics are intentionally leaked (raw new, never destroyed): a
+// py::object destructor running after Py_Finalize aborts the process.
 py::object& pickleDumps() {
-  static py::object f = py::module_::import("pickle").attr("dumps");
-  return f;
+  static py::object* f = new py::object(py::module_::import("pickle").attr("dumps"));
+  return *f;
 }
 py::object& pickleLoads() {
-  static py::object f = py::module_::import("pickle").attr("loads");
-  return f;
+  static py::object* f = new py::object(py::module_::import("pickle").attr("loads"));
+  return *f;
 }
 
 bool isNumpyArray(py::handle h) {
-  static py::object ndarrayType = []() -> py::object {
+  static py::object* ndarrayType = []() -> py::object* {
     try {
-      return py::module_::import("numpy").attr("ndarray");
+      return new py::object(py::module_::import("numpy").attr("ndarray"));
     } catch (...) {
-      return py::object();
+      return new py::object();
     }
   }();
-  return ndarrayType && py::isinstance(h, ndarrayType);
+  return ndarrayType->ptr() && py::isinstance(h, *ndarrayType);
 }
 
 }  // namespace
@@ -92,10 +94,11 @@ void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors
   } else if (isNumpyArray(obj)) {
     // Ride out-of-band as a tensor view when possible.
     try {
-      static py::object fromNumpy = py::module_::import("torch").attr("from_numpy");
+      static py::object* fromNumpy =
+          new py::object(py::module_::import("torch").attr("from_numpy"));  // leaked, see above
       py::object ascontig =
           py::module_::import("numpy").attr("ascontiguousarray")(py::reinterpret_borrow<py::object>(obj));
-      py::object t = fromNumpy(ascontig);
+      py::object t = (*fromNumpy)(ascontig);
       w.u8(tNdarray);
       w.u32(static_cast<uint32_t>(tensors.size()));
       tensors.push_back(THPVariable_Unpack(t.ptr()));
