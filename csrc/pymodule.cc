@@ -8,6 +8,7 @@
 
 #include <atomic>
 
+#include "accumulator.h"
 #include "pybits.h"
 #include "rpc.h"
 #include "serde.h"
@@ -351,6 +352,138 @@ class GroupWrapper {
   std::shared_ptr<Group> group_;
 };
 
+// --------------------------------------------------------- Accumulator
+
+class AccumulatorWrapper {
+ public:
+  AccumulatorWrapper(const std::string& name, py::object parameters, py::object buffers,
+                     py::object group) {
+    std::vector<at::Tensor> ps, bs;
+    for (auto h : parameters) ps.push_back(py::cast<at::Tensor>(h));
+    for (auto h : buffers) bs.push_back(py::cast<at::Tensor>(h));
+    if (group.is_none()) {
+      ownRpc_ = Rpc::create();
+      ownRpc_->setName(name + "-" + randomUid().substr(0, 8));
+      registerRpc(ownRpc_);
+      group_ = Group::create(ownRpc_, name + "_group");
+    } else {
+      group_ = py::cast<GroupWrapper&>(group).group();
+    }
+    acc_ = Accumulator::create(name, std::move(ps), std::move(bs), group_);
+  }
+
+  void connect(const std::string& addr) {
+    py::gil_scoped_release rel;
+    acc_->connect(addr);
+  }
+  void update() {
+    py::gil_scoped_release rel;
+    acc_->update();
+  }
+  bool connected() {
+    py::gil_scoped_release rel;
+    return acc_->connected();
+  }
+  bool wantsState() {
+    py::gil_scoped_release rel;
+    return acc_->wantsState();
+  }
+  bool hasNewState() {
+    py::gil_scoped_release rel;
+    return acc_->hasNewState();
+  }
+  void setState(py::object state) {
+    std::vector<at::Tensor> tensors;
+    std::string payload = serializeObject(state, tensors);
+    py::gil_scoped_release rel;
+    acc_->setState(std::move(payload), std::move(tensors));
+  }
+  py::object state() {
+    std::pair<std::string, std::vector<at::Tensor>> st;
+    {
+      py::gil_scoped_release rel;
+      st = acc_->state();
+    }
+    return deserializeObject(st.first, st.second);
+  }
+  bool wantsGradients() {
+    py::gil_scoped_release rel;
+    return acc_->wantsGradients();
+  }
+  bool hasGradients() {
+    py::gil_scoped_release rel;
+    return acc_->hasGradients();
+  }
+  void skipGradients() {
+    py::gil_scoped_release rel;
+    acc_->skipGradients();
+  }
+  void reduceGradients(int64_t batchSize) {
+    py::gil_scoped_release rel;
+    acc_->reduceGradients(batchSize);
+  }
+  void zeroGradients() {
+    py::gil_scoped_release rel;
+    acc_->zeroGradients();
+  }
+  int64_t modelVersion() {
+    py::gil_scoped_release rel;
+    return acc_->modelVersion();
+  }
+  void setModelVersion(int64_t v) {
+    py::gil_scoped_release rel;
+    acc_->setModelVersion(v);
+  }
+  std::string getLeader() {
+    py::gil_scoped_release rel;
+    return acc_->getLeader();
+  }
+  bool isLeader() {
+    py::gil_scoped_release rel;
+    return acc_->isLeader();
+  }
+  py::dict gradientStats() {
+    std::unordered_map<std::string, int64_t> s;
+    {
+      py::gil_scoped_release rel;
+      s = acc_->gradientStats();
+    }
+    py::dict d;
+    for (auto& [k, v] : s) d[py::str(k)] = v;
+    return d;
+  }
+  void setVirtualBatchSize(int64_t n) { acc_->setVirtualBatchSize(n); }
+  void setParallelGradients(int64_t n) { acc_->setParallelGradients(n); }
+  std::string debugState() {
+    py::gil_scoped_release rel;
+    return acc_->debugState();
+  }
+
+  void setLocalReduceHook(py::object fn) {
+    if (fn.is_none()) {
+      acc_->setLocalReduceHook(nullptr);
+      return;
+    }
+    auto g = std::make_shared<PyGuard>(fn);
+    acc_->setLocalReduceHook([g](at::Tensor& t) -> std::function<bool()> {
+      if (!pyAlive()) throw RpcError("interpreter shutting down");
+      py::gil_scoped_acquire gil;
+      py::object poll = g->obj(py::cast(t));
+      auto pg = std::make_shared<PyGuard>(poll);
+      return [pg]() -> bool {
+        if (!pyAlive()) return true;
+        py::gil_scoped_acquire gil;
+        return py::cast<bool>(pg->obj());
+      };
+    });
+  }
+
+ private:
+  Accumulator::Ptr acc_;
+  std::shared_ptr<Group> group_;
+  RpcPtr ownRpc_;
+};
+
 // -------------------------------------------------------------- module
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -448,6 +581,31 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("set_sort_order", &GroupWrapper::setSortOrder)
       .def("all_reduce", &GroupWrapper::allReduce, py::arg("name"), py::arg("value"),
            py::arg("op") = py::none());
+
+  py::class_<AccumulatorWrapper>(m, "Accumulator")
+      .def(py::init<const std::string&, py::object, py::object, py::object>(), py::arg("name"),
+           py::arg("parameters"), py::arg("buffers"), py::arg("group") = py::none())
+      .def("connect", &AccumulatorWrapper::connect, py::arg("address"))
+      .def("update", &AccumulatorWrapper::update)
+      .def("connected", &AccumulatorWrapper::connected)
+      .def("wants_state", &AccumulatorWrapper::wantsState)
+      .def("has_new_state", &AccumulatorWrapper::hasNewState)
+      .def("set_state", &AccumulatorWrapper::setState, py::arg("state"))
+      .def("state", &AccumulatorWrapper::state)
+      .def("wants_gradients", &AccumulatorWrapper::wantsGradients)
+      .def("has_gradients", &AccumulatorWrapper::hasGradients)
+      .def("skip_gradients", &AccumulatorWrapper::skipGradients)
+      .def("reduce_gradients", &AccumulatorWrapper::reduceGradients, py::arg("batch_size"))
+      .def("zero_gradients", &AccumulatorWrapper::zeroGradients)
+      .def("model_version", &AccumulatorWrapper::modelVersion)
+      .def("set_model_version", &AccumulatorWrapper::setModelVersion, py::arg("n"))
+      .def("get_leader", &AccumulatorWrapper::getLeader)
+      .def("is_leader", &AccumulatorWrapper::isLeader)
+      .def("get_gradient_stats", &AccumulatorWrapper::gradientStats)
+      .def("set_virtual_batch_size", &AccumulatorWrapper::setVirtualBatchSize, py::arg("n"))
+      .def("set_parallel_gradients", &AccumulatorWrapper::setParallelGradients, py::arg("n"))
+      .def("set_local_reduce_hook", &AccumulatorWrapper::setLocalReduceHook, py::arg("hook"))
+      .def("debug_state", &AccumulatorWrapper::debugState);
 }
 
 }  // namespace mrl

@@ -450,10 +450,14 @@ void Group::completeOp(const OpKey& key, AllReduceOpPtr op, ReduceValue* v, cons
     ops_.erase(key);
   }
   if (op->done) {
+    // Always deliver on a scheduler thread: an op can complete inline from
+    // inside allReduce() (single-member group), and callers (Accumulator)
+    // hold their own locks across allReduce().
     if (v) {
-      op->done(v, nullptr);
+      auto val = std::make_shared<ReduceValue>(std::move(*v));
+      globalScheduler().run([done = std::move(op->done), val] { done(val.get(), nullptr); });
     } else {
-      op->done(nullptr, &err);
+      globalScheduler().run([done = std::move(op->done), err] { done(nullptr, &err); });
     }
   }
 }

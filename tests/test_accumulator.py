@@ -1,0 +1,281 @@
+"""Accumulator tests: leader election, model sync, gradient reduction.
+
+Covers the reference's Accumulator protocol (examples/a2c.py usage pattern)
+on our implementation: in-process multi-peer over the RPC tree, and
+multi-process with the torch.distributed (gloo on CPU / RCCL on MI355X)
+collective data plane.
+"""
+import os
+import time
+
+import pytest
+import torch
+
+import moolib_amd
+
+
+def make_model(seed):
+    g = torch.Generator().manual_seed(seed)
+    w = torch.randn(4, 3, generator=g).requires_grad_()
+    b = torch.randn(4, generator=g).requires_grad_()
+    buf = torch.randn(2, generator=g)
+    return [w, b], [buf]
+
+
+class Peer:
+    def __init__(self, i, addr, group_name="accgroup"):
+        self.rpc = moolib_amd.Rpc()
+        self.rpc.set_name("peer%d" % i)
+        self.rpc.set_timeout(5)
+        self.group = moolib_amd.Group(self.rpc, group_name)
+        self.group.set_timeout(5)
+        self.params, self.buffers = make_model(seed=100 + i)
+        self.acc = moolib_amd.Accumulator(
+            "acc", self.params, self.buffers, group=self.group
+        )
+        self.acc.connect(addr)
+        self.state_set = 0
+        self.state_got = None
+
+    def pump_once(self):
+        self.acc.update()
+        if self.acc.wants_state():
+            self.acc.set_state({"opt": ("adam", self.rpc.get_name())})
+            self.state_set += 1
+        if self.acc.has_new_state():
+            self.state_got = self.acc.state()
+
+
+class AccCluster:
+    def __init__(self, n, virtual_batch_size=None):
+        self.broker_rpc = moolib_amd.Rpc()
+        self.broker_rpc.set_name("broker")
+        self.broker = moolib_amd.Broker(self.broker_rpc)
+        self.addr = self.broker_rpc.listen("127.0.0.1:0")[0]
+        self.peers = [Peer(i, self.addr) for i in range(n)]
+        if virtual_batch_size:
+            for p in self.peers:
+                p.acc.set_virtual_batch_size(virtual_batch_size)
+
+    def pump(self, cond, deadline=25):
+        t0 = time.time()
+        while time.time() - t0 < deadline:
+            self.broker.update()
+            for p in self.peers:
+                p.pump_once()
+            if cond():
+                return True
+            time.sleep(0.02)
+        return False
+
+    def wait_connected(self):
+        ok = self.pump(lambda: all(p.acc.connected() for p in self.peers))
+        assert ok, "accumulators did not connect: %s" % [
+            p.acc.debug_state() for p in self.peers
+        ]
+
+
+class TestAccumulator:
+    def test_connect_elect_and_model_sync(self):
+        c = AccCluster(3)
+        c.wait_connected()
+        leaders = set(p.acc.get_leader() for p in c.peers)
+        assert len(leaders) == 1
+        leader_name = leaders.pop()
+        n_leaders = sum(p.acc.is_leader() for p in c.peers)
+        assert n_leaders == 1
+        leader = next(p for p in c.peers if p.acc.is_leader())
+        assert leader.rpc.get_name() == leader_name
+        # Non-leaders adopted the leader's weights...
+        for p in c.peers:
+            for lp, pp in zip(leader.params, p.params):
+                assert torch.allclose(lp, pp)
+            for lb, pb in zip(leader.buffers, p.buffers):
+                assert torch.allclose(lb, pb)
+        # ...and received the leader's user state.
+        ok = c.pump(
+            lambda: all(p.acc.is_leader() or p.state_got is not None for p in c.peers)
+        )
+        assert ok
+        for p in c.peers:
+            if not p.acc.is_leader():
+                assert p.state_got == {"opt": ("adam", leader_name)}
+
+    def test_gradient_reduction(self):
+        c = AccCluster(2)
+        c.wait_connected()
+        grads = []
+        for i, p in enumerate(c.peers):
+            g = [torch.full_like(t, float(i + 1)) for t in p.params]
+            grads.append(g)
+
+        contributed = [False, False]
+        stepped = [False, False]
+
+        def step():
+            c.broker.update()
+            for i, p in enumerate(c.peers):
+                p.pump_once()
+                if p.acc.wants_gradients() and not contributed[i]:
+                    for t, g in zip(p.params, grads[i]):
+                        t.grad = g.clone()
+                    p.acc.reduce_gradients(8)
+                    contributed[i] = True
+                if p.acc.has_gradients():
+                    stats = p.acc.get_gradient_stats()
+                    assert stats["batch_size"] == 16
+                    assert stats["num_gradients"] == 2
+                    # average of per-peer gradients
+                    for t, g0, g1 in zip(p.params, grads[0], grads[1]):
+                        assert torch.allclose(t.grad, (g0 + g1) / 2)
+                    p.acc.zero_gradients()
+                    stepped[i] = True
+
+        t0 = time.time()
+        while not all(stepped) and time.time() - t0 < 25:
+            step()
+            time.sleep(0.01)
+        assert all(stepped), [p.acc.debug_state() for p in c.peers]
+
+    def test_virtual_batch_and_skip(self):
+        c = AccCluster(2, virtual_batch_size=32)
+        c.wait_connected()
+        # Peer 0 contributes batch 16 twice; peer 1 always skips.
+        contributions = [0]
+        done = [False]
+
+        def step():
+            c.broker.update()
+            p0, p1 = c.peers
+            p0.pump_once()
+            p1.pump_once()
+            if p0.acc.wants_gradients():
+                if contributions[0] < 2:
+                    for t in p0.params:
+                        t.grad = torch.ones_like(t)
+                    p0.acc.reduce_gradients(16)
+                    contributions[0] += 1
+                else:
+                    p0.acc.skip_gradients()
+            if p1.acc.wants_gradients():
+                p1.acc.skip_gradients()
+            if p0.acc.has_gradients():
+                stats = p0.acc.get_gradient_stats()
+                assert stats["batch_size"] == 32
+                assert stats["num_gradients"] == 2
+                assert stats["num_skipped"] >= 1
+                # two accumulated unit-gradients from peer0, num_gradients=2
+                for t in p0.params:
+                    assert torch.allclose(t.grad, torch.ones_like(t))
+                p0.acc.zero_gradients()
+                done[0] = True
+
+        t0 = time.time()
+        while not done[0] and time.time() - t0 < 25:
+            step()
+            time.sleep(0.01)
+        assert done[0], [p.acc.debug_state() for p in c.peers]
+
+    def test_model_version_increments(self):
+        c = AccCluster(1)
+        c.wait_connected()
+        p = c.peers[0]
+        v0 = p.acc.model_version()
+        done = []
+        t0 = time.time()
+        while not done and time.time() - t0 < 20:
+            c.broker.update()
+            p.pump_once()
+            if p.acc.wants_gradients():
+                for t in p.params:
+                    t.grad = torch.ones_like(t)
+                p.acc.reduce_gradients(1)
+            if p.acc.has_gradients():
+                p.acc.zero_gradients()
+                done.append(1)
+            time.sleep(0.01)
+        assert done
+        assert p.acc.model_version() == v0 + 1
+
+
+def _dist_worker(rank, world, port, results_dir):
+    import torch.distributed as dist
+
+    import moolib_amd
+    from moolib_amd import parallel
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    broker = None
+    if rank == 0:
+        broker_rpc = moolib_amd.Rpc()
+        broker_rpc.set_name("broker")
+        broker = moolib_amd.Broker(broker_rpc)
+        broker_rpc.listen("127.0.0.1:%d" % (port + 1))
+
+    rpc = moolib_amd.Rpc()
+    rpc.set_name("rank%d" % rank)
+    rpc.set_timeout(10)
+    group = moolib_amd.Group(rpc, "distgroup")
+    group.set_timeout(10)
+    params, buffers = make_model(seed=37 + rank)
+    acc = moolib_amd.Accumulator("acc", params, buffers, group=group)
+    acc.connect("127.0.0.1:%d" % (port + 1))
+    parallel.install_collective_backend(acc)
+
+    t0 = time.time()
+    applied = False
+    contributed = False
+    while time.time() - t0 < 60 and not applied:
+        if broker:
+            broker.update()
+        acc.update()
+        if acc.wants_state():
+            acc.set_state({"opt": rank})
+        if acc.has_new_state():
+            acc.state()
+        if acc.connected() and len(group.members()) == world:
+            if acc.wants_gradients() and not contributed:
+                for t in params:
+                    t.grad = torch.full_like(t, float(rank + 1))
+                acc.reduce_gradients(4)
+                contributed = True
+            if acc.has_gradients():
+                stats = acc.get_gradient_stats()
+                expect = sum(r + 1 for r in range(world)) / world
+                for t in params:
+                    assert torch.allclose(
+                        t.grad, torch.full_like(t, expect)
+                    ), "rank %d wrong grads" % rank
+                assert stats["num_gradients"] == world
+                applied = True
+        time.sleep(0.005)
+
+    assert applied, "rank %d never applied gradients: %s" % (rank, acc.debug_state())
+    with open(os.path.join(results_dir, "ok%d" % rank), "w") as f:
+        f.write("ok")
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+class TestDistributedDataPlane:
+    def test_gloo_collective_hook(self, tmp_path):
+        """2 ranks, gloo backend: same code path bench.py uses with RCCL."""
+        import torch.multiprocessing as mp
+
+        port = 29000 + (os.getpid() % 500)
+        world = 2
+        ctx = mp.get_context("spawn")
+        procs = [
+            ctx.Process(target=_dist_worker, args=(r, world, port, str(tmp_path)))
+            for r in range(world)
+        ]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=120)
+            assert p.exitcode == 0, "worker failed with %s" % p.exitcode
+        for r in range(world):
+            assert (tmp_path / ("ok%d" % r)).exists()
