@@ -9,6 +9,7 @@
 #include <atomic>
 
 #include "accumulator.h"
+#include "batcher.h"
 #include "pybits.h"
 #include "rpc.h"
 #include "serde.h"
@@ -581,6 +582,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("set_sort_order", &GroupWrapper::setSortOrder)
       .def("all_reduce", &GroupWrapper::allReduce, py::arg("name"), py::arg("value"),
            py::arg("op") = py::none());
+
+  py::class_<Batcher>(m, "Batcher")
+      .def(py::init<int64_t, py::object, int64_t>(), py::arg("size"),
+           py::arg("device") = py::none(), py::arg("dim") = 0)
+      .def("stack", &Batcher::stack, py::arg("tensors"))
+      .def("cat", &Batcher::cat, py::arg("tensors"))
+      .def("empty", &Batcher::empty)
+      .def("size", &Batcher::size)
+      .def("get", &Batcher::get)
+      .def("_pop_future", &Batcher::popFuture);
 
   py::class_<AccumulatorWrapper>(m, "Accumulator")
       .def(py::init<const std::string&, py::object, py::object, py::object>(), py::arg("name"),

@@ -76,6 +76,9 @@ Queue.__await__ = _queue_await
 Queue.__iter__ = _queue_await
 Queue.get = _queue_get
 
+_core.Batcher.__await__ = _queue_await
+_core.Batcher.__iter__ = _queue_await
+
 
 # Rpc with batched defines --------------------------------------------------
 
