@@ -10,6 +10,7 @@
 
 #include "accumulator.h"
 #include "batcher.h"
+#include "envpool.h"
 #include "pybits.h"
 #include "rpc.h"
 #include "serde.h"
@@ -582,6 +583,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("set_sort_order", &GroupWrapper::setSortOrder)
       .def("all_reduce", &GroupWrapper::allReduce, py::arg("name"), py::arg("value"),
            py::arg("op") = py::none());
+
+  py::class_<EnvStepperFuture>(m, "EnvStepperFuture")
+      .def("result", &EnvStepperFuture::result);
+
+  py::class_<EnvPool>(m, "EnvPool")
+      .def(py::init<py::object, int, int, int, int64_t>(), py::arg("create_env"),
+           py::arg("num_processes"), py::arg("batch_size"), py::arg("num_batches"),
+           py::arg("shared_memory_bytes") = 0)
+      .def("step", &EnvPool::step, py::arg("batch_index"), py::arg("action"))
+      .def("running", &EnvPool::running)
+      .def("num_workers_alive", &EnvPool::numWorkersAlive);
 
   py::class_<Batcher>(m, "Batcher")
       .def(py::init<int64_t, py::object, int64_t>(), py::arg("size"),

@@ -216,10 +216,39 @@ try:
 except ImportError:  # pragma: no cover — during staged bring-up
     pass
 
-try:
-    from ._core import EnvPool, EnvRunner, EnvStepper, EnvStepperFuture  # noqa: F401
-except ImportError:  # pragma: no cover
-    pass
+from ._core import EnvPool, EnvStepperFuture  # noqa: F401
+
+
+class EnvStepper:
+    """Client handle for stepping an EnvPool's batches.
+
+    API parity with the reference's EnvStepper (src/env.h:456-490); in this
+    implementation EnvPool itself is steppable, so this is a thin view.
+    """
+
+    def __init__(self, pool):
+        self._pool = pool
+
+    def step(self, batch_index, action):
+        return self._pool.step(batch_index, action)
+
+
+class EnvRunner:
+    """API-parity shim for the reference's EnvRunner (src/env.h:363-405).
+
+    The reference uses EnvRunner to host env workers reached over RPC; in
+    this implementation workers are forked by EnvPool directly, so the
+    runner only reports liveness.
+    """
+
+    def __init__(self, pool=None):
+        self._pool = pool
+
+    def start(self):
+        return None
+
+    def running(self):
+        return self._pool.running() if self._pool is not None else False
 
 
 atexit.register(_core._shutdown_all)

@@ -1,0 +1,118 @@
+"""EnvPool tests with a deterministic toy env (no gym dependency).
+
+Mirrors the reference's test/unit/test_envpool.py strategy: validate
+double-buffered stepping semantics, auto-reset behavior, and field shapes.
+"""
+import numpy as np
+import pytest
+import torch
+
+import moolib_amd
+
+
+class CountingEnv:
+    """obs = [count, last_action]; episode ends after 5 steps; reward = action."""
+
+    def __init__(self):
+        self.count = 0
+
+    def reset(self):
+        self.count = 0
+        return np.array([self.count, -1.0], dtype=np.float32)
+
+    def step(self, action):
+        self.count += 1
+        done = self.count >= 5
+        obs = np.array([self.count, float(action)], dtype=np.float32)
+        reward = float(action)
+        if done:
+            # the pool auto-resets and returns the new episode's first obs
+            pass
+        return obs, reward, done, {}
+
+
+class DictObsEnv:
+    def __init__(self):
+        self.t = 0
+
+    def reset(self):
+        self.t = 0
+        return {"a": np.zeros((2, 3), dtype=np.float32), "b": np.int64(self.t)}
+
+    def step(self, action):
+        self.t += 1
+        obs = {"a": np.full((2, 3), float(self.t), dtype=np.float32), "b": np.int64(self.t)}
+        return obs, 1.0, self.t >= 3, {}
+
+
+class TestEnvPool:
+    def test_basic_stepping(self):
+        pool = moolib_amd.EnvPool(CountingEnv, num_processes=2, batch_size=4, num_batches=1)
+        actions = torch.zeros(4, dtype=torch.int64)
+        # First step: reset -> count 0, reward 0, done False
+        obs = pool.step(0, actions).result()
+        assert set(obs.keys()) == {"state", "reward", "done"}
+        assert obs["state"].shape == (4, 2)
+        assert torch.equal(obs["state"][:, 0], torch.zeros(4))
+        assert torch.equal(obs["reward"], torch.zeros(4))
+        assert not obs["done"].any()
+        # Steps 1..4: counting up, reward = action
+        for i in range(1, 5):
+            actions = torch.full((4,), i % 3, dtype=torch.int64)
+            obs = pool.step(0, actions).result()
+            if i < 5:
+                assert torch.equal(obs["state"][:, 0], torch.full((4,), float(i)))
+                assert torch.equal(obs["state"][:, 1], torch.full((4,), float(i % 3)))
+                assert torch.allclose(obs["reward"], torch.full((4,), float(i % 3)))
+                assert not obs["done"].any()
+        # Step 5: done; obs is the auto-reset initial obs
+        obs = pool.step(0, torch.ones(4, dtype=torch.int64)).result()
+        assert obs["done"].all()
+        assert torch.equal(obs["state"][:, 0], torch.zeros(4))
+        assert torch.allclose(obs["reward"], torch.ones(4))
+        # After reset: counting starts again
+        obs = pool.step(0, torch.zeros(4, dtype=torch.int64)).result()
+        assert not obs["done"].any()
+        assert torch.equal(obs["state"][:, 0], torch.ones(4))
+
+    def test_double_buffering(self):
+        pool = moolib_amd.EnvPool(CountingEnv, num_processes=2, batch_size=3, num_batches=2)
+        f0 = pool.step(0, torch.zeros(3, dtype=torch.int64))
+        f1 = pool.step(1, torch.zeros(3, dtype=torch.int64))
+        o0, o1 = f0.result(), f1.result()
+        assert torch.equal(o0["state"][:, 0], torch.zeros(3))
+        assert torch.equal(o1["state"][:, 0], torch.zeros(3))
+        # batches advance independently
+        pool.step(0, torch.zeros(3, dtype=torch.int64)).result()
+        o0b = pool.step(0, torch.zeros(3, dtype=torch.int64)).result()
+        o1b = pool.step(1, torch.zeros(3, dtype=torch.int64)).result()
+        assert torch.equal(o0b["state"][:, 0], torch.full((3,), 2.0))
+        assert torch.equal(o1b["state"][:, 0], torch.full((3,), 1.0))
+
+    def test_dict_obs(self):
+        pool = moolib_amd.EnvPool(DictObsEnv, num_processes=1, batch_size=2, num_batches=1)
+        obs = pool.step(0, torch.zeros(2, dtype=torch.int64)).result()
+        assert set(obs.keys()) == {"a", "b", "reward", "done"}
+        assert obs["a"].shape == (2, 2, 3)
+        assert obs["a"].dtype == torch.float32
+        assert obs["b"].dtype == torch.int64
+        obs = pool.step(0, torch.zeros(2, dtype=torch.int64)).result()
+        assert torch.allclose(obs["a"], torch.ones(2, 2, 3))
+        assert torch.equal(obs["b"], torch.ones(2, dtype=torch.int64))
+
+    def test_worker_error_surfaces(self):
+        class BadEnv:
+            def reset(self):
+                raise ValueError("bad env is bad")
+
+            def step(self, a):
+                return None
+
+        pool = moolib_amd.EnvPool(BadEnv, num_processes=1, batch_size=1, num_batches=1)
+        with pytest.raises(RuntimeError, match="bad env is bad|died|failed"):
+            pool.step(0, torch.zeros(1, dtype=torch.int64)).result()
+
+    def test_running(self):
+        pool = moolib_amd.EnvPool(CountingEnv, num_processes=2, batch_size=2, num_batches=1)
+        assert pool.running()
+        assert pool.num_workers_alive() == 2
