@@ -1,0 +1,189 @@
+#!/usr/bin/env python
+"""Flagship benchmark: IMPALA Atari-ResNet env frames/sec (whole node).
+
+Measures the metric BASELINE.json names — env frames/sec consumed for
+training by the IMPALA/V-trace loop on the Atari deep-ResNet model at the
+reference benchmark config (actor_batch_size 128 x 2 batches, 10 actor
+processes, learn batch 32, unroll 20, virtual_batch_size 32, Adam 6e-4) on
+synthetic 84x84x4 uint8 frames with random-init weights.
+
+Single process per GPU; launched for N>1 by torch.distributed.run
+(one rank per GPU over RCCL); the Accumulator's gradient bucket reduces via
+dist.all_reduce (RCCL over xGMI) while membership/counts run on the moolib
+RPC plane. Weak scaling: per-GPU actor+learner work is fixed as N grows.
+
+One JSON line is printed by rank 0 at the end (driver contract).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30, help="timed optimizer steps")
+    ap.add_argument("--warmup", type=int, default=10, help="untimed warmup optimizer steps")
+    ap.add_argument("--use-lstm", action="store_true")
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--actor-batch-size", type=int, default=128)
+    ap.add_argument("--num-actor-batches", type=int, default=2)
+    ap.add_argument("--num-actor-cpus", type=int, default=10)
+    ap.add_argument("--batch-size", type=int, default=32)
+    ap.add_argument("--unroll-length", type=int, default=20)
+    ap.add_argument("--virtual-batch-size", type=int, default=32)
+    ap.add_argument("--max-seconds", type=float, default=1800.0)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    # ---- 1. Fork env workers FIRST (clean pre-CUDA, pre-thread processes).
+    import moolib_amd
+    from moolib_amd.envs import SyntheticAtariEnv
+
+    num_actions = 18
+    envs = moolib_amd.EnvPool(
+        lambda: SyntheticAtariEnv(num_actions=num_actions, mean_episode_len=1000),
+        num_processes=args.num_actor_cpus,
+        batch_size=args.actor_batch_size,
+        num_batches=args.num_actor_batches,
+    )
+
+    # ---- 2. Now CUDA / torch.distributed.
+    import torch
+
+    use_cuda = torch.cuda.is_available()
+    if args.device:
+        device = args.device
+    elif use_cuda:
+        device = "cuda:%d" % local_rank
+    else:
+        device = "cpu"
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.init_process_group("nccl" if use_cuda else "gloo", rank=rank, world_size=world)
+
+    # ---- 3. Control plane: rank 0 hosts the broker in-process.
+    master_port = int(os.environ.get("MASTER_PORT", "0"))
+    broker_port = master_port + 7 if master_port else 0
+    broker = None
+    if rank == 0:
+        broker_rpc = moolib_amd.Rpc()
+        broker_rpc.set_name("broker")
+        broker = moolib_amd.Broker(broker_rpc)
+        bound = broker_rpc.listen("127.0.0.1:%d" % broker_port)
+        broker_addr = [a for a in bound if a.startswith("tcp://127")][0]
+    if world > 1:
+        obj = [broker_addr] if rank == 0 else [None]
+        dist.broadcast_object_list(obj, src=0)
+        broker_addr = obj[0]
+    elif rank == 0:
+        pass  # broker_addr already set
+
+    from moolib_amd.impala import ImpalaConfig, ImpalaPeer
+
+    cfg = ImpalaConfig(
+        num_actions=num_actions,
+        actor_batch_size=args.actor_batch_size,
+        num_actor_batches=args.num_actor_batches,
+        num_actor_cpus=args.num_actor_cpus,
+        batch_size=args.batch_size,
+        unroll_length=args.unroll_length,
+        virtual_batch_size=args.virtual_batch_size,
+        device=device,
+        use_lstm=args.use_lstm,
+        connect=broker_addr,
+        local_name="rank%d" % rank,
+        group_name="bench",
+        lr_schedule=False,
+    )
+    peer = ImpalaPeer(
+        cfg,
+        envs=envs,
+        use_collective_backend=world > 1,
+        broker=broker,
+    )
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    def run_steps(n, budget):
+        t_end = time.time() + budget
+        done = 0
+        while done < n:
+            if time.time() > t_end:
+                raise TimeoutError("benchmark stalled: %d/%d optimizer steps" % (done, n))
+            ev = peer.step_once()
+            if ev == "optimize":
+                done += 1
+
+    # ---- warmup ----
+    run_steps(args.warmup, args.max_seconds / 2)
+
+    # ---- timed region ----
+    vbs_stat = peer.stats["virtual_batch_size"]
+    v0 = vbs_stat.value  # running sum of global batch sizes over rounds
+    barrier_sync()
+    t0 = time.time()
+    run_steps(args.steps, args.max_seconds)
+    barrier_sync()
+    elapsed = time.time() - t0
+    frames = (vbs_stat.value - v0) * args.unroll_length  # global frames consumed
+
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if use_cuda:
+            t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        dtype = "bf16" if peer.autocast else "fp32"
+        result = {
+            "metric": "env frames/sec (whole node) IMPALA Atari-ResNet",
+            "value": frames / elapsed,
+            "unit": "frames/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": {
+                "model": "IMPALA Atari deep ResNet (16/32/32ch, FC3872-256%s)"
+                % ("+LSTM256" if args.use_lstm else ""),
+                "global_batch": args.batch_size * world,
+                "seq_len": args.unroll_length,
+                "parallelism": "dp%d" % world,
+                "actor_batch_size": args.actor_batch_size,
+                "num_actor_batches": args.num_actor_batches,
+                "num_actor_cpus": args.num_actor_cpus,
+                "virtual_batch_size": args.virtual_batch_size,
+                "num_actions": num_actions,
+                "frame": "84x84x4 uint8",
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if dist is not None:
+        dist.barrier()
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -73,7 +73,11 @@ Batcher::Batcher(int64_t size, py::object device, int64_t dim)
     : size_(size), dim_(dim) {
   if (size_ <= 0) throw RpcError("batcher: size must be positive");
   if (!device.is_none()) {
-    device_ = py::cast<at::Device>(device);  // torch's caster accepts torch.device and str
+    if (py::isinstance<py::str>(device)) {
+      device_ = at::Device(py::cast<std::string>(device));
+    } else {
+      device_ = py::cast<at::Device>(device);
+    }
   }
 }
 

@@ -1,0 +1,124 @@
+"""IMPALA deep ResNet for Atari (Espeholt et al. 2018, "deep" variant).
+
+Architecture parity with the reference model (examples/atari/models.py:9-153)
+— this exact config is what BASELINE.json's headline benchmark names:
+3 blocks of (3x3 conv -> 3x3/2 maxpool -> 2 residual blocks) with 16/32/32
+channels, FC 3872->256, one-hot prev action + clipped reward appended,
+optional LSTM(256), policy + baseline heads, multinomial sampling inside
+forward.
+
+MI355X notes: forward runs under bf16 autocast from the IMPALA loop (MFMA
+conv/GEMM via MIOpen/hipBLASLt); the actor path is hipGraph-captured by the
+learner loop (fixed [1, B] shapes), which removes the ~70-kernel launch
+overhead that dominates small-CNN inference.
+"""
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+
+class ResidualBlock(nn.Module):
+    def __init__(self, ch):
+        super().__init__()
+        self.conv0 = nn.Conv2d(ch, ch, 3, stride=1, padding=1)
+        self.conv1 = nn.Conv2d(ch, ch, 3, stride=1, padding=1)
+
+    def forward(self, x):
+        y = self.conv0(F.relu(x))
+        y = self.conv1(F.relu(y))
+        return x + y
+
+
+class ConvSection(nn.Module):
+    def __init__(self, in_ch, out_ch):
+        super().__init__()
+        self.conv = nn.Conv2d(in_ch, out_ch, 3, stride=1, padding=1)
+        self.pool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.res0 = ResidualBlock(out_ch)
+        self.res1 = ResidualBlock(out_ch)
+
+    def forward(self, x):
+        x = self.pool(self.conv(x))
+        x = self.res0(x)
+        x = self.res1(x)
+        return x
+
+
+class AtariNet(nn.Module):
+    def __init__(self, num_actions=18, input_channels=4, use_lstm=False):
+        super().__init__()
+        self.num_actions = num_actions
+        self.use_lstm = use_lstm
+
+        chans = [16, 32, 32]
+        sections = []
+        ch_in = input_channels
+        for ch in chans:
+            sections.append(ConvSection(ch_in, ch))
+            ch_in = ch
+        self.sections = nn.ModuleList(sections)
+
+        self.fc = nn.Linear(3872, 256)
+        core_in = self.fc.out_features + num_actions + 1
+        if use_lstm:
+            self.core = nn.LSTM(core_in, 256, num_layers=1)
+            core_in = 256
+        self.policy = nn.Linear(core_in, num_actions)
+        self.baseline = nn.Linear(core_in, 1)
+
+    def initial_state(self, batch_size=1):
+        if not self.use_lstm:
+            return tuple()
+        return tuple(
+            torch.zeros(self.core.num_layers, batch_size, self.core.hidden_size)
+            for _ in range(2)
+        )
+
+    def forward(self, inputs, core_state=None):
+        x = inputs["state"]
+        reward = inputs["reward"]
+        T, B = x.shape[:2]
+        x = x.flatten(0, 1).float().mul_(1.0 / 255.0)
+
+        for s in self.sections:
+            x = s(x)
+        x = F.relu(x)
+        x = x.view(T * B, -1)
+        x = F.relu(self.fc(x))
+
+        prev_action_onehot = F.one_hot(
+            inputs["prev_action"].view(T * B).to(torch.int64), self.num_actions
+        ).to(x.dtype)
+        clipped_reward = reward.view(T * B, 1).clamp(-1, 1).to(x.dtype)
+        core_input = torch.cat([x, clipped_reward, prev_action_onehot], dim=-1)
+
+        if self.use_lstm:
+            done = inputs["done"]
+            core_input = core_input.view(T, B, -1)
+            notdone = (~done).to(core_input.dtype)
+            outputs = []
+            for t in range(T):
+                nd = notdone[t].view(1, -1, 1)
+                core_state = tuple(nd * s for s in core_state)
+                out_t, core_state = self.core(core_input[t : t + 1], core_state)
+                outputs.append(out_t)
+            core_output = torch.cat(outputs).flatten(0, 1)
+        else:
+            core_output = core_input
+
+        policy_logits = self.policy(core_output).float()
+        baseline = self.baseline(core_output).float()
+        action = torch.multinomial(F.softmax(policy_logits, dim=-1), num_samples=1)
+
+        out = dict(
+            policy_logits=policy_logits.view(T, B, self.num_actions),
+            baseline=baseline.view(T, B),
+            action=action.view(T, B),
+        )
+        return out, core_state
+
+
+def create_model(num_actions=18, use_lstm=False, device="cpu"):
+    model = AtariNet(num_actions=num_actions, use_lstm=use_lstm)
+    model.to(device)
+    return model

@@ -1,0 +1,152 @@
+"""GPU (MI355X) tests: HIP kernel numerics vs plain fp32 torch references,
+plus device paths of batcher/accumulator/impala."""
+import time
+
+import pytest
+import torch
+
+import moolib_amd
+
+gpu = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+
+
+@gpu
+@requires_gpu
+class TestVtraceKernel:
+    def test_matches_cpu_reference(self):
+        from moolib_amd.ops import vtrace
+
+        torch.manual_seed(0)
+        for T, B in [(20, 32), (1, 1), (80, 640), (20, 1024)]:
+            log_rhos = torch.randn(T, B) * 0.4
+            discounts = (torch.rand(T, B) > 0.05).float() * 0.99
+            rewards = torch.randn(T, B)
+            values = torch.randn(T, B)
+            bootstrap = torch.randn(B)
+            cpu = vtrace.from_importance_weights(log_rhos, discounts, rewards, values, bootstrap)
+            dev = vtrace.from_importance_weights(
+                log_rhos.cuda(), discounts.cuda(), rewards.cuda(), values.cuda(), bootstrap.cuda()
+            )
+            assert torch.allclose(dev.vs.cpu(), cpu.vs, atol=1e-4), (T, B)
+            assert torch.allclose(dev.pg_advantages.cpu(), cpu.pg_advantages, atol=1e-4)
+
+    def test_kernel_used_on_gpu(self):
+        """The fused kernel must actually run (no silent eager fallback)."""
+        from moolib_amd import _kernels
+
+        out = _kernels.vtrace_from_log_rhos(
+            torch.zeros(4, 8, device="cuda"),
+            torch.full((4, 8), 0.9, device="cuda"),
+            torch.ones(4, 8, device="cuda"),
+            torch.zeros(4, 8, device="cuda"),
+            torch.zeros(8, device="cuda"),
+            1.0,
+            1.0,
+        )
+        torch.cuda.synchronize()
+        assert out[0].shape == (4, 8)
+
+
+@gpu
+@requires_gpu
+class TestFusedLoss:
+    def test_matches_eager(self):
+        from moolib_amd.ops import fused_loss, losses
+
+        torch.manual_seed(1)
+        T, B, A = 20, 32, 18
+        for ec, bc in [(0.0006, 0.5), (0.01, 1.0)]:
+            logits = (torch.randn(T, B, A, device="cuda") * 2).requires_grad_()
+            baseline = torch.randn(T, B, device="cuda").requires_grad_()
+            actions = torch.randint(0, A, (T, B), device="cuda")
+            pg_adv = torch.randn(T, B, device="cuda")
+            vs = torch.randn(T, B, device="cuda")
+
+            total = fused_loss.impala_total_loss(logits, baseline, actions, pg_adv, vs, ec, bc)
+            total.backward()
+            g_logits, g_baseline = logits.grad.clone(), baseline.grad.clone()
+
+            logits2 = logits.detach().clone().requires_grad_()
+            baseline2 = baseline.detach().clone().requires_grad_()
+            pg = losses.policy_gradient_loss(logits2, actions, pg_adv)
+            bl = bc * losses.baseline_loss(vs - baseline2)
+            en = ec * losses.entropy_loss(logits2)
+            ref = pg + bl + en
+            ref.backward()
+
+            assert torch.allclose(total.detach(), ref.detach(), atol=1e-4)
+            assert torch.allclose(g_logits, logits2.grad, atol=1e-5)
+            assert torch.allclose(g_baseline, baseline2.grad, atol=1e-5)
+
+
+@gpu
+@requires_gpu
+class TestDevicePaths:
+    def test_batcher_on_gpu(self):
+        b = moolib_amd.Batcher(4, "cuda:0", dim=1)
+        x = torch.randn(5, 8, 3)
+        b.cat({"x": x})
+        o1 = b.get()
+        assert o1["x"].is_cuda
+        assert torch.allclose(o1["x"].cpu(), x[:, :4])
+
+    def test_accumulator_gpu_bucket(self):
+        """Single-peer accumulator with CUDA params: flat bucket on device,
+        RPC-tree staging path."""
+        broker_rpc = moolib_amd.Rpc()
+        broker_rpc.set_name("broker")
+        broker = moolib_amd.Broker(broker_rpc)
+        addr = broker_rpc.listen("127.0.0.1:0")[0]
+        params = [torch.randn(64, 32, device="cuda").requires_grad_()]
+        rpc = moolib_amd.Rpc()
+        rpc.set_name("p0")
+        group = moolib_amd.Group(rpc, "gputest")
+        acc = moolib_amd.Accumulator("acc", params, [], group=group)
+        acc.connect(addr)
+        t0 = time.time()
+        applied = False
+        while time.time() - t0 < 30 and not applied:
+            broker.update()
+            acc.update()
+            if acc.wants_state():
+                acc.set_state({})
+            if acc.connected() and acc.wants_gradients():
+                params[0].grad = torch.full_like(params[0], 2.0)
+                acc.reduce_gradients(1)
+            if acc.has_gradients():
+                assert torch.allclose(params[0].grad, torch.full_like(params[0], 2.0))
+                acc.zero_gradients()
+                applied = True
+            time.sleep(0.005)
+        assert applied
+
+    def test_impala_gpu_short(self):
+        from moolib_amd.envs import SyntheticAtariEnv
+        from moolib_amd.impala import ImpalaConfig, ImpalaPeer
+
+        broker_rpc = moolib_amd.Rpc()
+        broker_rpc.set_name("broker")
+        broker = moolib_amd.Broker(broker_rpc)
+        addr = broker_rpc.listen("127.0.0.1:0")[0]
+        cfg = ImpalaConfig(
+            num_actions=6,
+            actor_batch_size=16,
+            num_actor_batches=2,
+            num_actor_cpus=2,
+            batch_size=8,
+            unroll_length=5,
+            virtual_batch_size=8,
+            device="cuda:0",
+            connect=addr,
+            total_steps=1e6,
+        )
+        peer = ImpalaPeer(cfg, lambda: SyntheticAtariEnv(num_actions=6), broker=broker)
+        t0 = time.time()
+        opt_steps = 0
+        while opt_steps < 3 and time.time() - t0 < 120:
+            if peer.step_once() == "optimize":
+                opt_steps += 1
+        assert opt_steps >= 3
+        assert peer.stats["env_train_steps"].result() >= 3 * 5 * 8
