@@ -327,6 +327,8 @@ void Accumulator::startGradReduceLocked() {
                         self->flat_.copy_(result, /*non_blocking=*/true);
                         self->applyGradResultLocked(self->flat_);
                       } else {
+                        // (Group::completeOp already cloned the tensors, so
+                        // in-place mutation here is safe.)
                         self->applyGradResultLocked(result);
                       }
                     });

@@ -455,6 +455,10 @@ void Group::completeOp(const OpKey& key, AllReduceOpPtr op, ReduceValue* v, cons
     // hold their own locks across allReduce().
     if (v) {
       auto val = std::make_shared<ReduceValue>(std::move(*v));
+      // Clone tensors before delivery: the same storages may still be
+      // referenced by in-flight zero-copy sends down the tree, and the
+      // consumer is free to mutate the result in place.
+      for (auto& t : val->tensors) t = t.clone();
       globalScheduler().run([done = std::move(op->done), val] { done(val.get(), nullptr); });
     } else {
       globalScheduler().run([done = std::move(op->done), err] { done(nullptr, &err); });
