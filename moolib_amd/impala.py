@@ -303,7 +303,7 @@ class ImpalaPeer:
             )
         actor_outputs = nest.map(lambda t: t.squeeze(0), actor_outputs)
         action = actor_outputs["action"]
-        env_state.update(cpu_env_outputs, action.cpu(), self.stats)
+        env_state.update(cpu_env_outputs, action, self.stats)
         del cpu_env_outputs  # aliases shm; next step() overwrites it
         env_state.future = self.envs.step(cur, action)
         self.stats["env_act_steps"] += action.numel()
