@@ -35,6 +35,7 @@ def main():
     ap.add_argument("--unroll-length", type=int, default=20)
     ap.add_argument("--virtual-batch-size", type=int, default=32)
     ap.add_argument("--max-seconds", type=float, default=1800.0)
+    ap.add_argument("--breakdown", action="store_true", help="print per-phase wall time")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -131,6 +132,9 @@ def main():
 
     # ---- warmup ----
     run_steps(args.warmup, args.max_seconds / 2)
+    if args.breakdown:
+        peer.profile = True
+        peer.phase_times = {}
 
     # ---- timed region ----
     vbs_stat = peer.stats["virtual_batch_size"]
@@ -148,6 +152,15 @@ def main():
             t = t.cuda()
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
+
+    if args.breakdown and rank == 0:
+        total = sum(peer.phase_times.values())
+        print(
+            "phase breakdown (%.3fs measured / %.3fs elapsed):" % (total, elapsed),
+            file=sys.stderr,
+        )
+        for k, v in sorted(peer.phase_times.items(), key=lambda kv: -kv[1]):
+            print("  %-16s %8.3fs  %5.1f%%" % (k, v, 100 * v / elapsed), file=sys.stderr)
 
     if rank == 0:
         dtype = "bf16" if peer.autocast else "fp32"

@@ -169,6 +169,19 @@ class ImpalaPeer:
         self.next_env_index = 0
         self.model_version = 0
         self.autocast = cfg.autocast_bf16 and torch.device(cfg.device).type == "cuda"
+        # Phase profiling (bench --breakdown): cumulative seconds per phase,
+        # with device sync at boundaries so GPU time is attributed correctly.
+        self.profile = False
+        self.phase_times = {}
+
+    def _t(self, name, t0):
+        if self.profile:
+            if torch.device(self.cfg.device).type == "cuda":
+                torch.cuda.synchronize()
+            t1 = time.perf_counter()
+            self.phase_times[name] = self.phase_times.get(name, 0.0) + (t1 - t0)
+            return t1
+        return t0
 
     # ------------------------------------------------------------ learning
 
@@ -263,16 +276,20 @@ class ImpalaPeer:
             time.sleep(0.05)
             return "idle"
 
+        t0 = time.perf_counter() if self.profile else 0.0
         if acc.has_gradients():
             gstats = acc.get_gradient_stats()
             self.stats["virtual_batch_size"] += gstats["batch_size"]
             self.stats["num_gradients"] += gstats["num_gradients"]
             self.step_optimizer()
             acc.zero_gradients()
+            self._t("optimize", t0)
             return "optimize"
         elif not self.learn_batcher.empty() and acc.wants_gradients():
             self.compute_gradients(self.learn_batcher.get())
+            t0 = self._t("learn_fwd_bwd", t0)
             acc.reduce_gradients(cfg.batch_size)
+            self._t("learn_reduce", t0)
             return "learn"
         else:
             if acc.wants_gradients():
@@ -285,14 +302,17 @@ class ImpalaPeer:
         cur = self.next_env_index
         self.next_env_index = (self.next_env_index + 1) % cfg.num_actor_batches
         env_state = self.env_states[cur]
+        t0 = time.perf_counter() if self.profile else 0.0
         if env_state.future is None:
             env_state.future = self.envs.step(cur, env_state.prev_action)
         cpu_env_outputs = env_state.future.result()
+        t0 = self._t("act_env_wait", t0)
 
         env_outputs = nest.map(
             lambda t: t.to(cfg.device, copy=True, non_blocking=True), cpu_env_outputs
         )
         env_outputs["prev_action"] = env_state.prev_action
+        t0 = self._t("act_h2d", t0)
         prev_core_state = env_state.core_state
         self.model.eval()
         with torch.no_grad(), torch.autocast(
@@ -303,10 +323,12 @@ class ImpalaPeer:
             )
         actor_outputs = nest.map(lambda t: t.squeeze(0), actor_outputs)
         action = actor_outputs["action"]
+        t0 = self._t("act_forward", t0)
         env_state.update(cpu_env_outputs, action, self.stats)
         del cpu_env_outputs  # aliases shm; next step() overwrites it
         env_state.future = self.envs.step(cur, action)
         self.stats["env_act_steps"] += action.numel()
+        t0 = self._t("act_stats_step", t0)
 
         last_data = {"env_outputs": env_outputs, "actor_outputs": actor_outputs}
         env_state.time_batcher.stack(last_data)
@@ -317,3 +339,4 @@ class ImpalaPeer:
             # Carry the last entry of the previous unroll into the next one.
             env_state.initial_core_state = prev_core_state
             env_state.time_batcher.stack(last_data)
+        self._t("act_batch", t0)
