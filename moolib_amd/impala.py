@@ -55,6 +55,10 @@ class ImpalaConfig:
     local_name: str = ""
     group_name: str = "impala"
     lr_schedule: bool = True
+    channels_last: bool = True   # NHWC convs (MIOpen direct, no transposes)
+    graph_actor: bool = True     # hipGraph-capture the actor forward
+    graph_learner: bool = True   # hipGraph-capture the learner fwd+bwd
+    pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
 
 
 class EnvBatchState:
@@ -70,9 +74,12 @@ class EnvBatchState:
         self.running_reward = torch.zeros(B)
         self.step_count = torch.zeros(B)
         self.time_batcher = moolib_amd.Batcher(cfg.unroll_length + 1, device)
+        self.pinned = {}  # field -> pinned staging tensor
 
     def update(self, env_outputs, action, stats):
-        self.prev_action = action
+        # prev_action is a persistent device buffer (stable address for
+        # hipGraph replay); copy instead of rebinding.
+        self.prev_action.copy_(action, non_blocking=True)
         self.running_reward += env_outputs["reward"]
         self.step_count += 1
         done = env_outputs["done"]
@@ -168,7 +175,22 @@ class ImpalaPeer:
         self.global_stats_accumulator = GlobalStatsAccumulator(self.group, make_stats())
         self.next_env_index = 0
         self.model_version = 0
-        self.autocast = cfg.autocast_bf16 and torch.device(cfg.device).type == "cuda"
+        self.is_cuda = torch.device(cfg.device).type == "cuda"
+        self.autocast = cfg.autocast_bf16 and self.is_cuda
+        if self.is_cuda and cfg.channels_last:
+            self.model.to(memory_format=torch.channels_last)
+        from moolib_amd.parallel.graphs import GraphedCall
+
+        self._actor_call = (
+            GraphedCall(self._actor_fn, warmup=3, name="actor_fwd")
+            if (self.is_cuda and cfg.graph_actor)
+            else self._actor_fn
+        )
+        self._learn_call = (
+            GraphedCall(self._learn_fn, warmup=3, name="learner_fwd_bwd")
+            if (self.is_cuda and cfg.graph_learner)
+            else self._learn_fn
+        )
         # Phase profiling (bench --breakdown): cumulative seconds per phase,
         # with device sync at boundaries so GPU time is attributed correctly.
         self.profile = False
@@ -185,15 +207,18 @@ class ImpalaPeer:
 
     # ------------------------------------------------------------ learning
 
-    def compute_gradients(self, data):
+    def _learn_fn(self, data):
+        """Forward + V-trace + fused loss + backward. hipGraph-capturable:
+        fixed shapes, writes gradients into the (stable) param.grad tensors."""
         cfg = self.cfg
         model = self.model
         env_outputs = data["env_outputs"]
         actor_outputs = data["actor_outputs"]
         initial_core_state = data["initial_core_state"]
-        model.train()
 
-        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=self.autocast):
+        with torch.autocast(
+            "cuda", dtype=torch.bfloat16, enabled=self.autocast, cache_enabled=False
+        ):
             learner_outputs, _ = model(env_outputs, initial_core_state)
 
         bootstrap_value = learner_outputs["baseline"][-1]
@@ -226,7 +251,23 @@ class ImpalaPeer:
             cfg.baseline_cost,
         )
         total.backward()
-        self.stats["env_train_steps"] += cfg.unroll_length * cfg.batch_size
+        return ()
+
+    def compute_gradients(self, data):
+        self.model.train()
+        self._learn_call(data)
+        self.stats["env_train_steps"] += self.cfg.unroll_length * self.cfg.batch_size
+
+    def _actor_fn(self, inputs):
+        """No-grad actor forward on [1, B]. hipGraph-capturable."""
+        env_outputs, core_state = inputs["env"], inputs["core"]
+        with torch.no_grad(), torch.autocast(
+            "cuda", dtype=torch.bfloat16, enabled=self.autocast, cache_enabled=False
+        ):
+            actor_outputs, core_out = self.model(
+                nest.map(lambda t: t.unsqueeze(0), env_outputs), core_state
+            )
+        return {"out": actor_outputs, "core": core_out if core_out else ()}
 
     def step_optimizer(self):
         cfg = self.cfg
@@ -308,20 +349,37 @@ class ImpalaPeer:
         cpu_env_outputs = env_state.future.result()
         t0 = self._t("act_env_wait", t0)
 
-        env_outputs = nest.map(
-            lambda t: t.to(cfg.device, copy=True, non_blocking=True), cpu_env_outputs
-        )
+        if self.is_cuda and cfg.pinned_staging:
+            # shm -> pinned -> HBM: the shm-backed views are pageable, so a
+            # direct .to() is a slow synchronous copy; the pinned bounce
+            # buffer makes the H2D leg an async DMA.
+            env_outputs = {}
+            for k, t in cpu_env_outputs.items():
+                pin = env_state.pinned.get(k)
+                if pin is None:
+                    pin = torch.empty_like(t).pin_memory()
+                    env_state.pinned[k] = pin
+                pin.copy_(t)
+                env_outputs[k] = pin.to(cfg.device, non_blocking=True)
+        else:
+            env_outputs = nest.map(
+                lambda t: t.to(cfg.device, copy=True, non_blocking=True), cpu_env_outputs
+            )
         env_outputs["prev_action"] = env_state.prev_action
         t0 = self._t("act_h2d", t0)
-        prev_core_state = env_state.core_state
+        use_lstm = cfg.use_lstm
+        prev_core_state = (
+            tuple(t.clone() for t in env_state.core_state) if use_lstm else tuple()
+        )
         self.model.eval()
-        with torch.no_grad(), torch.autocast(
-            "cuda", dtype=torch.bfloat16, enabled=self.autocast
-        ):
-            actor_outputs, env_state.core_state = self.model(
-                nest.map(lambda t: t.unsqueeze(0), env_outputs), env_state.core_state
-            )
-        actor_outputs = nest.map(lambda t: t.squeeze(0), actor_outputs)
+        res = self._actor_call({"env": env_outputs, "core": env_state.core_state})
+        actor_outputs = nest.map(lambda t: t.squeeze(0), res["out"])
+        if use_lstm:
+            # keep persistent core buffers (stable addresses for graph replay)
+            new_core = res["core"]
+            for dst, src in zip(env_state.core_state, new_core):
+                if dst.data_ptr() != src.data_ptr():
+                    dst.copy_(src, non_blocking=True)
         action = actor_outputs["action"]
         t0 = self._t("act_forward", t0)
         env_state.update(cpu_env_outputs, action, self.stats)

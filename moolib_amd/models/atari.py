@@ -79,11 +79,17 @@ class AtariNet(nn.Module):
         reward = inputs["reward"]
         T, B = x.shape[:2]
         x = x.flatten(0, 1).float().mul_(1.0 / 255.0)
+        if x.is_cuda:
+            # NHWC: MIOpen's bf16 igemm kernels are NHWC-native; NCHW input
+            # inserts a batched_transpose around every conv. (Note: the FC
+            # input ordering then differs from the CPU/NCHW path — a
+            # self-consistent permutation of learned features.)
+            x = x.contiguous(memory_format=torch.channels_last)
 
         for s in self.sections:
             x = s(x)
         x = F.relu(x)
-        x = x.view(T * B, -1)
+        x = x.reshape(T * B, -1)
         x = F.relu(self.fc(x))
 
         prev_action_onehot = F.one_hot(
