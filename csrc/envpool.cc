@@ -176,6 +176,13 @@ class EnvPoolImpl : public std::enable_shared_from_this<EnvPoolImpl> {
     return batchDicts_[b];
   }
 
+  at::Tensor dataRegion() {
+    // Whole obs/reward/done region as one uint8 view — the GPU path
+    // hipHostRegisters it once so shm->HBM copies run as async DMA.
+    return at::from_blob(base_ + dataOff_, {static_cast<int64_t>(segBytes_ - dataOff_)},
+                         at::TensorOptions().dtype(at::kByte));
+  }
+
   bool anyAlive() {
     for (pid_t p : pids_) {
       if (waitpid(p, nullptr, WNOHANG) == 0) return true;
@@ -436,6 +443,7 @@ EnvStepperFuture EnvPool::step(int batchIndex, py::object action) {
   return EnvStepperFuture(impl_, batchIndex);
 }
 
+at::Tensor EnvPool::sharedBuffer() { return impl_->dataRegion(); }
 bool EnvPool::running() { return impl_->anyAlive(); }
 int EnvPool::numWorkersAlive() { return impl_->aliveCount(); }
 

@@ -46,6 +46,7 @@ class EnvPool {
           int64_t sharedMemoryBytes);
   ~EnvPool();
   EnvStepperFuture step(int batchIndex, py::object action);
+  at::Tensor sharedBuffer();
   bool running();
   int numWorkersAlive();
 

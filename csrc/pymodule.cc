@@ -592,6 +592,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("num_processes"), py::arg("batch_size"), py::arg("num_batches"),
            py::arg("shared_memory_bytes") = 0)
       .def("step", &EnvPool::step, py::arg("batch_index"), py::arg("action"))
+      .def("shared_buffer", &EnvPool::sharedBuffer)
       .def("running", &EnvPool::running)
       .def("num_workers_alive", &EnvPool::numWorkersAlive);
 

@@ -192,8 +192,21 @@ std::vector<at::Tensor> impala_loss(at::Tensor logits, at::Tensor actions,
   return {loss_parts, grad_logits, grad_baseline};
 }
 
+// Pin an existing CPU range (e.g. the EnvPool shared-memory segment) so
+// tensor views into it qualify for async H2D DMA.
+void register_host_memory(at::Tensor t) {
+  TORCH_CHECK(t.device().is_cpu(), "register_host_memory: CPU tensor expected");
+  hipError_t err = hipHostRegister(t.data_ptr(), t.nbytes(), hipHostRegisterDefault);
+  if (err == hipErrorHostMemoryAlreadyRegistered) {
+    (void)hipGetLastError();
+    return;
+  }
+  TORCH_CHECK(err == hipSuccess, "hipHostRegister failed: ", hipGetErrorString(err));
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "moolib_amd gfx950 HIP kernels";
+  m.def("register_host_memory", &register_host_memory);
   m.def("vtrace_from_log_rhos", &vtrace_from_log_rhos, "fused V-trace scan (gfx950)");
   m.def("impala_loss", &impala_loss, "fused IMPALA loss fwd+grad (gfx950)");
 }

@@ -191,6 +191,15 @@ class ImpalaPeer:
             if (self.is_cuda and cfg.graph_learner)
             else self._learn_fn
         )
+        self._shm_registered = False
+        if self.is_cuda and cfg.pinned_staging:
+            try:
+                from moolib_amd import _kernels
+
+                _kernels.register_host_memory(self.envs.shared_buffer())
+                self._shm_registered = True
+            except Exception:  # noqa: BLE001
+                pass  # fall back to pinned bounce buffers
         # Phase profiling (bench --breakdown): cumulative seconds per phase,
         # with device sync at boundaries so GPU time is attributed correctly.
         self.profile = False
@@ -349,7 +358,12 @@ class ImpalaPeer:
         cpu_env_outputs = env_state.future.result()
         t0 = self._t("act_env_wait", t0)
 
-        if self.is_cuda and cfg.pinned_staging:
+        if self._shm_registered:
+            # shm region is hipHostRegistered: direct async DMA, no bounce.
+            env_outputs = {
+                k: t.to(cfg.device, non_blocking=True) for k, t in cpu_env_outputs.items()
+            }
+        elif self.is_cuda and cfg.pinned_staging:
             # shm -> pinned -> HBM: the shm-backed views are pageable, so a
             # direct .to() is a slow synchronous copy; the pinned bounce
             # buffer makes the H2D leg an async DMA.
