@@ -59,6 +59,7 @@ class ImpalaConfig:
     graph_actor: bool = True     # hipGraph-capture the actor forward
     graph_learner: bool = True   # hipGraph-capture the learner fwd+bwd
     pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
+    shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
 
 
 class EnvBatchState:
@@ -192,7 +193,7 @@ class ImpalaPeer:
             else self._learn_fn
         )
         self._shm_registered = False
-        if self.is_cuda and cfg.pinned_staging:
+        if self.is_cuda and cfg.shm_host_register:
             try:
                 from moolib_amd import _kernels
 
