@@ -15,6 +15,7 @@ MI355X design:
   - the actor forward (fixed [1, B] shape) is hipGraph-captured when
     enabled (cfg.graph_actor) to eliminate launch overhead.
 """
+import contextlib
 import dataclasses
 import time
 from typing import Any, Optional
@@ -60,6 +61,7 @@ class ImpalaConfig:
     graph_learner: bool = True   # hipGraph-capture the learner fwd+bwd
     pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
     shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
+    actor_side_stream: bool = True   # overlap actor H2D+forward with the learner stream
 
 
 class EnvBatchState:
@@ -180,6 +182,9 @@ class ImpalaPeer:
         self.autocast = cfg.autocast_bf16 and self.is_cuda
         if self.is_cuda and cfg.channels_last:
             self.model.to(memory_format=torch.channels_last)
+        self.actor_stream = (
+            torch.cuda.Stream() if (self.is_cuda and cfg.actor_side_stream) else None
+        )
         from moolib_amd.parallel.graphs import GraphedCall
 
         self._actor_call = (
@@ -334,9 +339,15 @@ class ImpalaPeer:
             self.stats["num_gradients"] += gstats["num_gradients"]
             self.step_optimizer()
             acc.zero_gradients()
+            if self.actor_stream is not None:
+                # next actor forwards must see the updated weights
+                self.actor_stream.wait_stream(torch.cuda.current_stream())
             self._t("optimize", t0)
             return "optimize"
         elif not self.learn_batcher.empty() and acc.wants_gradients():
+            if self.actor_stream is not None:
+                # learner consumes batches written on the actor stream
+                torch.cuda.current_stream().wait_stream(self.actor_stream)
             self.compute_gradients(self.learn_batcher.get())
             t0 = self._t("learn_fwd_bwd", t0)
             acc.reduce_gradients(cfg.batch_size)
@@ -358,6 +369,15 @@ class ImpalaPeer:
             env_state.future = self.envs.step(cur, env_state.prev_action)
         cpu_env_outputs = env_state.future.result()
         t0 = self._t("act_env_wait", t0)
+        stream_ctx = (
+            torch.cuda.stream(self.actor_stream)
+            if self.actor_stream is not None
+            else contextlib.nullcontext()
+        )
+        with stream_ctx:
+            self._act_body(cfg, cur, env_state, cpu_env_outputs, t0)
+
+    def _act_body(self, cfg, cur, env_state, cpu_env_outputs, t0):
 
         if self._shm_registered:
             # shm region is hipHostRegistered: direct async DMA, no bounce.
