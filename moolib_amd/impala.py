@@ -62,6 +62,7 @@ class ImpalaConfig:
     pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
     shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
     actor_side_stream: bool = True   # overlap actor H2D+forward with the learner stream
+    bf16_weights: bool = True        # forward on a bf16 shadow of the fp32 master weights
 
 
 class EnvBatchState:
@@ -72,7 +73,10 @@ class EnvBatchState:
         device = cfg.device
         self.prev_action = torch.zeros(B, dtype=torch.int64, device=device)
         self.future = None
-        self.core_state = tuple(s.to(device) for s in model.initial_state(batch_size=B))
+        dt = next(model.parameters()).dtype
+        self.core_state = tuple(
+            s.to(device=device, dtype=dt) for s in model.initial_state(batch_size=B)
+        )
         self.initial_core_state = self.core_state
         self.running_reward = torch.zeros(B)
         self.step_count = torch.zeros(B)
@@ -173,7 +177,6 @@ class ImpalaPeer:
             parallel.install_collective_backend(self.accumulator)
 
         self.learn_batcher = moolib_amd.Batcher(cfg.batch_size, cfg.device, dim=1)
-        self.env_states = [EnvBatchState(cfg, self.model) for _ in range(cfg.num_actor_batches)]
         self.stats = make_stats()
         self.global_stats_accumulator = GlobalStatsAccumulator(self.group, make_stats())
         self.next_env_index = 0
@@ -182,6 +185,28 @@ class ImpalaPeer:
         self.autocast = cfg.autocast_bf16 and self.is_cuda
         if self.is_cuda and cfg.channels_last:
             self.model.to(memory_format=torch.channels_last)
+        # bf16 shadow: forwards/backwards run on bf16 weights (no per-use
+        # autocast casts inside every graph replay); the fp32 master owns
+        # the optimizer, the accumulator and model sync. Gradients are cast
+        # bf16->fp32 once per learn; weights fp32->bf16 once per optimizer
+        # step.
+        self.bf16_shadow = self.autocast and cfg.bf16_weights
+        if self.bf16_shadow:
+            import copy as _copy
+
+            self.fwd_model = _copy.deepcopy(self.model).to(torch.bfloat16)
+            if cfg.channels_last:
+                self.fwd_model.to(memory_format=torch.channels_last)
+            self.autocast = False
+            self._master_params = [p for p in self.model.parameters() if p.requires_grad]
+            self._fwd_params = [p for p in self.fwd_model.parameters() if p.requires_grad]
+        else:
+            self.fwd_model = self.model
+            self._master_params = []
+            self._fwd_params = []
+        self.env_states = [
+            EnvBatchState(cfg, self.fwd_model) for _ in range(cfg.num_actor_batches)
+        ]
         self.actor_stream = (
             torch.cuda.Stream() if (self.is_cuda and cfg.actor_side_stream) else None
         )
@@ -222,11 +247,24 @@ class ImpalaPeer:
 
     # ------------------------------------------------------------ learning
 
+    def _sync_shadow(self):
+        if not self.bf16_shadow:
+            return
+        with torch.no_grad():
+            for pf, pb in zip(self._master_params, self._fwd_params):
+                pb.copy_(pf, non_blocking=True)
+            for bf, bb in zip(self.model.buffers(), self.fwd_model.buffers()):
+                bb.copy_(bf, non_blocking=True)
+
     def _learn_fn(self, data):
         """Forward + V-trace + fused loss + backward. hipGraph-capturable:
         fixed shapes, writes gradients into the (stable) param.grad tensors."""
         cfg = self.cfg
-        model = self.model
+        model = self.fwd_model
+        if self.bf16_shadow:
+            for pb in self._fwd_params:
+                if pb.grad is not None:
+                    pb.grad.zero_()
         env_outputs = data["env_outputs"]
         actor_outputs = data["actor_outputs"]
         initial_core_state = data["initial_core_state"]
@@ -266,10 +304,20 @@ class ImpalaPeer:
             cfg.baseline_cost,
         )
         total.backward()
+        if self.bf16_shadow:
+            # cast this batch's bf16 gradients onto the fp32 masters
+            with torch.no_grad():
+                for pf, pb in zip(self._master_params, self._fwd_params):
+                    if pb.grad is None:
+                        continue
+                    if pf.grad is None:
+                        pf.grad = pb.grad.float()
+                    else:
+                        pf.grad.copy_(pb.grad, non_blocking=True)
         return ()
 
     def compute_gradients(self, data):
-        self.model.train()
+        self.fwd_model.train()
         self._learn_call(data)
         self.stats["env_train_steps"] += self.cfg.unroll_length * self.cfg.batch_size
 
@@ -279,7 +327,7 @@ class ImpalaPeer:
         with torch.no_grad(), torch.autocast(
             "cuda", dtype=torch.bfloat16, enabled=self.autocast, cache_enabled=False
         ):
-            actor_outputs, core_out = self.model(
+            actor_outputs, core_out = self.fwd_model(
                 nest.map(lambda t: t.unsqueeze(0), env_outputs), core_state
             )
         return {"out": actor_outputs, "core": core_out if core_out else ()}
@@ -288,6 +336,7 @@ class ImpalaPeer:
         cfg = self.cfg
         norm = torch.nn.utils.clip_grad_norm_(self.model.parameters(), cfg.grad_norm_clipping)
         self.optimizer.step()
+        self._sync_shadow()
         if self.scheduler is not None:
             self.scheduler.step()
         self.model_version += 1
@@ -327,6 +376,7 @@ class ImpalaPeer:
             acc.set_state(self.save_state())
         if acc.has_new_state():
             self.load_state(acc.state())
+            self._sync_shadow()  # master params were replaced by the leader's
 
         if not acc.connected():
             time.sleep(0.05)
@@ -406,7 +456,7 @@ class ImpalaPeer:
         prev_core_state = (
             tuple(t.clone() for t in env_state.core_state) if use_lstm else tuple()
         )
-        self.model.eval()
+        self.fwd_model.eval()
         res = self._actor_call({"env": env_outputs, "core": env_state.core_state})
         actor_outputs = nest.map(lambda t: t.squeeze(0), res["out"])
         if use_lstm:

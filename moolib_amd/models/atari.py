@@ -78,7 +78,7 @@ class AtariNet(nn.Module):
         x = inputs["state"]
         reward = inputs["reward"]
         T, B = x.shape[:2]
-        x = x.flatten(0, 1).float().mul_(1.0 / 255.0)
+        x = x.flatten(0, 1).to(self.fc.weight.dtype).mul_(1.0 / 255.0)
         if x.is_cuda:
             # NHWC: MIOpen's bf16 igemm kernels are NHWC-native; NCHW input
             # inserts a batched_transpose around every conv. (Note: the FC
