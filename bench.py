@@ -163,7 +163,7 @@ def main():
             print("  %-16s %8.3fs  %5.1f%%" % (k, v, 100 * v / elapsed), file=sys.stderr)
 
     if rank == 0:
-        dtype = "bf16" if peer.autocast else "fp32"
+        dtype = "bf16" if (peer.autocast or getattr(peer, "bf16_shadow", False)) else "fp32"
         result = {
             "metric": "env frames/sec (whole node) IMPALA Atari-ResNet",
             "value": frames / elapsed,

@@ -118,6 +118,9 @@ __global__ void impala_loss_kernel(const float* __restrict__ logits,
   h = waveReduceSum(h);
 
   int64_t a = actions[row];
+  // Defensive clamp: an out-of-range action (corrupted input) must not
+  // fault the GPU; training numerics elsewhere will surface the bug.
+  if (a < 0 || a >= A) a = 0;
   float adv = pg_advantages[row];
   float logp_a = lrow[a] - logz;
 

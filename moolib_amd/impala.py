@@ -62,7 +62,7 @@ class ImpalaConfig:
     pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
     shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
     actor_side_stream: bool = True   # overlap actor H2D+forward with the learner stream
-    bf16_weights: bool = True        # forward on a bf16 shadow of the fp32 master weights
+    bf16_weights: bool = False       # forward on a bf16 shadow of the fp32 master weights (measured neutral vs autocast)
 
 
 class EnvBatchState:
@@ -210,6 +210,7 @@ class ImpalaPeer:
         self.actor_stream = (
             torch.cuda.Stream() if (self.is_cuda and cfg.actor_side_stream) else None
         )
+        self._batch_events = []
         from moolib_amd.parallel.graphs import GraphedCall
 
         self._actor_call = (
@@ -395,10 +396,20 @@ class ImpalaPeer:
             self._t("optimize", t0)
             return "optimize"
         elif not self.learn_batcher.empty() and acc.wants_gradients():
+            data = self.learn_batcher.get()
             if self.actor_stream is not None:
-                # learner consumes batches written on the actor stream
-                torch.cuda.current_stream().wait_stream(self.actor_stream)
-            self.compute_gradients(self.learn_batcher.get())
+                # wait only for the event recorded when THIS batch completed,
+                # and tell the allocator the tensors are now used on the
+                # learner stream (they were written on the actor stream).
+                if self._batch_events:
+                    torch.cuda.current_stream().wait_event(self._batch_events.pop(0))
+                else:
+                    torch.cuda.current_stream().wait_stream(self.actor_stream)
+                cur = torch.cuda.current_stream()
+                for t in nest.flatten(data):
+                    if isinstance(t, torch.Tensor) and t.is_cuda:
+                        t.record_stream(cur)
+            self.compute_gradients(data)
             t0 = self._t("learn_fwd_bwd", t0)
             acc.reduce_gradients(cfg.batch_size)
             self._t("learn_reduce", t0)
@@ -478,7 +489,13 @@ class ImpalaPeer:
         if not env_state.time_batcher.empty():
             data = env_state.time_batcher.get()
             data["initial_core_state"] = env_state.initial_core_state
+            n_before = self.learn_batcher.size()
             self.learn_batcher.cat(data)
+            if self.actor_stream is not None:
+                for _ in range(self.learn_batcher.size() - n_before):
+                    ev = torch.cuda.Event()
+                    ev.record(self.actor_stream)
+                    self._batch_events.append(ev)
             # Carry the last entry of the previous unroll into the next one.
             env_state.initial_core_state = prev_core_state
             env_state.time_batcher.stack(last_data)
