@@ -272,9 +272,9 @@ class BrokerWrapper {
     broker_ = std::make_unique<Broker>(rpc_);
   }
   void setName(const std::string& n) { rpc_->setName(n); }
-  void listen(const std::string& addr) {
+  std::vector<std::string> listen(const std::string& addr) {
     py::gil_scoped_release rel;
-    rpc_->listen(addr);
+    return rpc_->listen(addr);
   }
   void update() { broker_->update(); }
   RpcPtr rpc() { return rpc_; }
