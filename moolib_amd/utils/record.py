@@ -17,7 +17,7 @@ def log_to_file(path="logs.tsv", **fields):
         os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
         writeheader = not os.path.exists(path)
         f = open(path, "a", buffering=1)
-        writer = csv.DictWriter(f, list(fields.keys()), delimiter="\t")
+        writer = csv.DictWriter(f, list(fields.keys()), delimiter="\t", extrasaction="ignore", restval="")
         if writeheader:
             writer.writeheader()
         _writers[path] = state = (f, writer)
