@@ -35,6 +35,7 @@ def main():
     ap.add_argument("--unroll-length", type=int, default=20)
     ap.add_argument("--virtual-batch-size", type=int, default=32)
     ap.add_argument("--max-seconds", type=float, default=1800.0)
+    ap.add_argument("--backend", default=None, help="torch.distributed backend override")
     ap.add_argument("--breakdown", action="store_true", help="print per-phase wall time")
     args = ap.parse_args()
 
@@ -69,9 +70,14 @@ def main():
 
     dist = None
     if world > 1:
+        import datetime
+
         import torch.distributed as dist
 
-        dist.init_process_group("nccl" if use_cuda else "gloo", rank=rank, world_size=world)
+        backend = args.backend or ("nccl" if use_cuda else "gloo")
+        dist.init_process_group(
+            backend, rank=rank, world_size=world, timeout=datetime.timedelta(seconds=300)
+        )
 
     # ---- 3. Control plane: rank 0 hosts the broker in-process.
     master_port = int(os.environ.get("MASTER_PORT", "0"))
