@@ -66,7 +66,8 @@ def main():
     else:
         device = "cpu"
     if use_cuda:
-        torch.cuda.set_device(local_rank)
+        dev_index = torch.device(device).index
+        torch.cuda.set_device(dev_index if dev_index is not None else local_rank)
 
     dist = None
     if world > 1:
