@@ -62,6 +62,7 @@ class ImpalaConfig:
     pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
     shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
     actor_side_stream: bool = True   # overlap actor H2D+forward with the learner stream
+    max_learn_backlog: int = 4       # stop acting when this many learn batches are queued
     bf16_weights: bool = False       # forward on a bf16 shadow of the fp32 master weights (measured neutral vs autocast)
 
 
@@ -417,6 +418,12 @@ class ImpalaPeer:
         else:
             if acc.wants_gradients():
                 acc.skip_gradients()
+            if self.learn_batcher.size() >= cfg.max_learn_backlog:
+                # Backpressure: the learner is behind (slow peer / stalled
+                # round) — unbounded acting would grow the ready queue (and
+                # HBM use) without limit.
+                time.sleep(0.001)
+                return "throttle"
             self.act_once()
             return "act"
 
