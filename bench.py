@@ -45,6 +45,9 @@ def main():
 
     # ---- 1. Fork env workers FIRST (clean pre-CUDA, pre-thread processes).
     import moolib_amd
+
+    if os.environ.get("MOOLIB_AMD_LOG"):
+        moolib_amd.set_log_level(os.environ["MOOLIB_AMD_LOG"])
     from moolib_amd.envs import SyntheticAtariEnv
 
     num_actions = 18
@@ -132,6 +135,19 @@ def main():
         done = 0
         while done < n:
             if time.time() > t_end:
+                print(
+                    "STALL rank %d: %s | group %s active=%s | batcher=%d | events=%s"
+                    % (
+                        rank,
+                        peer.accumulator.debug_state(),
+                        peer.group.members(),
+                        peer.group.active(),
+                        peer.learn_batcher.size(),
+                        {k: v.result() for k, v in peer.stats.items()},
+                    ),
+                    file=sys.stderr,
+                    flush=True,
+                )
                 raise TimeoutError("benchmark stalled: %d/%d optimizer steps" % (done, n))
             ev = peer.step_once()
             if ev == "optimize":
