@@ -116,6 +116,7 @@ def main():
         local_name="rank%d" % rank,
         group_name="bench",
         lr_schedule=False,
+        group_timeout=60.0,  # benchmark ranks never churn; avoid spurious eviction
     )
     peer = ImpalaPeer(
         cfg,

@@ -124,8 +124,12 @@ void Accumulator::resetLocked(const char* why) {
   newBatch_ = newGrads_ = newSkipped_ = 0;
   totBatch_ = totGrads_ = totSkipped_ = 0;
   hasGradients_ = false;
-  hookPoll_ = nullptr;
-  {
+  hookStartPending_ = false;
+  if (hookPoll_) {
+    // Never abandon an in-flight collective: the peers' call sequences must
+    // stay aligned. Keep polling it to completion (drain), then discard.
+    hookAbandoned_ = true;
+  } else {
     NoGrad ng;
     flat_.zero_();
   }
@@ -156,6 +160,7 @@ void Accumulator::startElectionLocked() {
                                      self->name_.c_str(), err->c_str());
                         self->phase_ = Phase::inactive;
                         self->syncSeen_ = 0;
+                        self->group_->requestResync();
                         return;
                       }
                       self->leader_ = rv->leaderName;
@@ -178,6 +183,23 @@ void Accumulator::update() {
   uint64_t sid = group_->syncId();
 
   std::unique_lock<std::mutex> lk(mu_);
+  // Drain an abandoned hook collective regardless of phase: the peers'
+  // collective call sequences must stay aligned, so it is polled to
+  // completion and its (mixed-round) result discarded.
+  if (hookPoll_ && hookAbandoned_) {
+    bool done = true;
+    try {
+      done = hookPoll_();
+    } catch (const std::exception& e) {
+      MRL_LOG_ERROR("abandoned hook drain failed: %s", e.what());
+    }
+    if (done) {
+      hookPoll_ = nullptr;
+      hookAbandoned_ = false;
+      NoGrad ng;
+      flat_.zero_();
+    }
+  }
   if (!active) {
     if (phase_ != Phase::inactive) resetLocked("group inactive");
     phase_ = Phase::inactive;
@@ -229,16 +251,39 @@ void Accumulator::update() {
   }
 
   if (phase_ == Phase::running) {
+    if (gradPhase_ == GradPhase::reducing &&
+        secondsSince(gradPhaseStarted_) > 120.0) {
+      MRL_LOG_ERROR("accumulator '%s': gradient reduce wedged >120s; forcing resync",
+                    name_.c_str());
+      if (hookPoll_) hookAbandoned_ = true;
+      phase_ = Phase::inactive;
+      syncSeen_ = 0;
+      group_->requestResync();
+      return;
+    }
     if (gradPhase_ == GradPhase::wantDecision && decided_) {
       startCountRoundLocked();
-    } else if (gradPhase_ == GradPhase::reducing && hookPoll_) {
+    } else if (gradPhase_ == GradPhase::reducing && hookStartPending_ && !hookPoll_) {
+      hookStartPending_ = false;
+      try {
+        hookPoll_ = hook_(flat_);
+      } catch (const std::exception& e) {
+        MRL_LOG_ERROR("local reduce hook failed: %s", e.what());
+        phase_ = Phase::inactive;
+        syncSeen_ = 0;
+        group_->requestResync();
+        return;
+      }
+    } else if (gradPhase_ == GradPhase::reducing && hookPoll_ && !hookAbandoned_) {
       bool done = false;
       try {
         done = hookPoll_();
       } catch (const std::exception& e) {
         MRL_LOG_ERROR("local reduce hook poll failed: %s", e.what());
+        hookPoll_ = nullptr;
         phase_ = Phase::inactive;
         syncSeen_ = 0;
+        group_->requestResync();
         return;
       }
       if (done) {
@@ -272,6 +317,7 @@ void Accumulator::startCountRoundLocked() {
                                      self->name_.c_str(), err->c_str());
                         self->phase_ = Phase::inactive;
                         self->syncSeen_ = 0;
+                        self->group_->requestResync();
                         return;
                       }
                       self->totBatch_ += rv->batchSize;
@@ -290,6 +336,12 @@ void Accumulator::startGradReduceLocked() {
   gradPhase_ = GradPhase::reducing;
   gradPhaseStarted_ = now();
   if (hook_) {
+    if (hookPoll_) {
+      // Previous (abandoned) collective still draining: start ours once it
+      // completes (update() watches hookStartPending_).
+      hookStartPending_ = true;
+      return;
+    }
     // MI355X fast path: in-place sum over the fixed torch.distributed world
     // (RCCL over xGMI). Peers that skipped contribute zeros.
     try {
@@ -298,6 +350,7 @@ void Accumulator::startGradReduceLocked() {
       MRL_LOG_ERROR("local reduce hook failed: %s", e.what());
       phase_ = Phase::inactive;
       syncSeen_ = 0;
+      group_->requestResync();
     }
     return;
   }
@@ -319,6 +372,7 @@ void Accumulator::startGradReduceLocked() {
                                      self->name_.c_str(), err->c_str());
                         self->phase_ = Phase::inactive;
                         self->syncSeen_ = 0;
+                        self->group_->requestResync();
                         return;
                       }
                       NoGrad ng;

@@ -124,6 +124,8 @@ class Accumulator : public std::enable_shared_from_this<Accumulator> {
   int64_t statBatch_ = 0, statGrads_ = 0, statSkipped_ = 0;  // of the applied result
   LocalReduceHook hook_;
   std::function<bool()> hookPoll_;
+  bool hookAbandoned_ = false;     // in-flight collective whose round we left
+  bool hookStartPending_ = false;  // next collective waits for the drain
   TimePoint gradPhaseStarted_{};
 };
 

@@ -89,6 +89,7 @@ void ReduceValue::fold(ReduceValue& src, const std::function<void(ReduceValue&, 
 
 namespace {
 constexpr const char* kBrokerPing = "__mrl_broker_ping";
+constexpr const char* kBrokerResync = "__mrl_broker_resync";
 }
 
 void Broker::GroupState::resort() {
@@ -102,6 +103,28 @@ void Broker::GroupState::resort() {
 }
 
 Broker::Broker(RpcPtr rpc) : rpc_(std::move(rpc)) {
+  rpc_->define(kBrokerResync, [this](Frame f, const std::string& fromPeer, RespondFn respond) {
+    try {
+      WireReader r(f.payload);
+      std::string group(r.str());
+      std::lock_guard<std::mutex> lk(mu_);
+      // Rate-limit: concurrent failure reports need only one epoch bump.
+      if (secondsSince(lastForcedResync_) > 0.5) {
+        lastForcedResync_ = now();
+        auto it = groups_.find(group);
+        if (it != groups_.end()) {
+          it->second.syncId++;
+          it->second.dirty = true;
+          it->second.resort();
+          MRL_LOG_INFO("broker: forced resync of group '%s' by '%s' (syncId now %llu)",
+                       group.c_str(), fromPeer.c_str(), (unsigned long long)it->second.syncId);
+        }
+      }
+      respond("", {}, false);
+    } catch (const std::exception& e) {
+      respond(e.what(), {}, true);
+    }
+  });
   rpc_->define(kBrokerPing, [this](Frame f, const std::string& fromPeer, RespondFn respond) {
     try {
       WireReader r(f.payload);
@@ -149,7 +172,10 @@ Broker::Broker(RpcPtr rpc) : rpc_(std::move(rpc)) {
 }
 
 Broker::~Broker() {
-  if (rpc_ && !rpc_->isShutdown()) rpc_->undefine(kBrokerPing);
+  if (rpc_ && !rpc_->isShutdown()) {
+    rpc_->undefine(kBrokerPing);
+    rpc_->undefine(kBrokerResync);
+  }
 }
 
 void Broker::setName(const std::string& n) { rpc_->setName(n); }
@@ -339,6 +365,13 @@ void Group::update() {
   }
   std::string err = "allreduce timed out";
   for (auto& op : timedOut) op->done(nullptr, &err);
+}
+
+void Group::requestResync() {
+  WireWriter w;
+  w.str(name_);
+  rpc_->sendRequest(brokerName_, kBrokerResync, std::move(w.out), {},
+                    [](Frame*, const std::string*) {}, 10.0);
 }
 
 void Group::allReduce(const std::string& opName, ReduceValue value, PyFold fold, ReduceDone done) {

@@ -60,6 +60,7 @@ class Broker {
   void update();  // evict stale members, bump syncIds
 
  private:
+  TimePoint lastForcedResync_{};
   struct Member {
     TimePoint lastPing{};
     int64_t sortOrder = 0;
@@ -132,6 +133,10 @@ class Group : public std::enable_shared_from_this<Group> {
   // Start an allreduce. All members must start ops with the same name in the
   // same order. done runs on a scheduler thread (or inline).
   void allReduce(const std::string& opName, ReduceValue value, PyFold fold, ReduceDone done);
+
+  // Ask the broker to bump this group's syncId so every member resets to a
+  // fresh epoch (re-aligns allreduce sequence counters after a failure).
+  void requestResync();
 
   RpcPtr rpc() { return rpc_; }
   std::string myName() { return rpc_->getName(); }

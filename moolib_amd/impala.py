@@ -63,6 +63,7 @@ class ImpalaConfig:
     shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
     actor_side_stream: bool = True   # overlap actor H2D+forward with the learner stream
     max_learn_backlog: int = 4       # stop acting when this many learn batches are queued
+    group_timeout: float = 10.0      # broker eviction / allreduce timeout (seconds)
     bf16_weights: bool = False       # forward on a bf16 shadow of the fp32 master weights (measured neutral vs autocast)
 
 
@@ -169,6 +170,7 @@ class ImpalaPeer:
         self.rpc.set_name(cfg.local_name or ("peer-" + moolib_amd.create_uid()[:8]))
         self.rpc.set_timeout(20)
         self.group = moolib_amd.Group(self.rpc, cfg.group_name)
+        self.group.set_timeout(cfg.group_timeout)
         self.accumulator = moolib_amd.Accumulator(
             "model", self.model.parameters(), self.model.buffers(), group=self.group
         )
