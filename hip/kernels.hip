@@ -143,6 +143,84 @@ __global__ void impala_loss_kernel(const float* __restrict__ logits,
   }
 }
 
+// ------------------------------------------------- NHWC max pooling 3x3/2
+//
+// The IMPALA ResNet's three pools (84->42->21->11, k3 s2 p1). torch's NHWC
+// maxpool backward scatters with atomics (~186 us on [672,16,42,42] grads);
+// this pair stores a window index in the forward and GATHERS in the
+// backward (each input element checks its <=4 covering windows) — no
+// atomics, bandwidth-bound.
+
+template <typename T>
+__global__ void maxpool3x3s2_fwd_kernel(const T* __restrict__ in, T* __restrict__ out,
+                                        uint8_t* __restrict__ idx, int N, int H, int W, int C,
+                                        int OH, int OW) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)N * OH * OW * C;
+  if (tid >= total) return;
+  int c = tid % C;
+  int64_t t = tid / C;
+  int ow = t % OW;
+  t /= OW;
+  int oh = t % OH;
+  int n = t / OH;
+
+  float best = -INFINITY;
+  int besti = 0;
+#pragma unroll
+  for (int kh = 0; kh < 3; ++kh) {
+    int ih = oh * 2 - 1 + kh;
+    if (ih < 0 || ih >= H) continue;
+#pragma unroll
+    for (int kw = 0; kw < 3; ++kw) {
+      int iw = ow * 2 - 1 + kw;
+      if (iw < 0 || iw >= W) continue;
+      float v = (float)in[(((int64_t)n * H + ih) * W + iw) * C + c];
+      if (v > best) {
+        best = v;
+        besti = kh * 3 + kw;
+      }
+    }
+  }
+  out[tid] = (T)best;
+  idx[tid] = (uint8_t)besti;
+}
+
+template <typename T>
+__global__ void maxpool3x3s2_bwd_kernel(const T* __restrict__ gout,
+                                        const uint8_t* __restrict__ idx, T* __restrict__ gin,
+                                        int N, int H, int W, int C, int OH, int OW) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)N * H * W * C;
+  if (tid >= total) return;
+  int c = tid % C;
+  int64_t t = tid / C;
+  int iw = t % W;
+  t /= W;
+  int ih = t % H;
+  int n = t / H;
+
+  float acc = 0.f;
+  // windows (oh, ow) covering (ih, iw): oh*2-1 <= ih <= oh*2+1
+  int oh_lo = (ih - 1 + 1) / 2;  // ceil((ih-1)/2) for ih>=1; 0 for ih==0
+  if (ih == 0) oh_lo = 0;
+  int oh_hi = (ih + 1) / 2;
+  int ow_lo = (iw == 0) ? 0 : (iw - 1 + 1) / 2;
+  if (iw == 0) ow_lo = 0;
+  int ow_hi = (iw + 1) / 2;
+  for (int oh = oh_lo; oh <= oh_hi && oh < OH; ++oh) {
+    int kh = ih - (oh * 2 - 1);
+    if (kh < 0 || kh > 2) continue;
+    for (int ow = ow_lo; ow <= ow_hi && ow < OW; ++ow) {
+      int kw = iw - (ow * 2 - 1);
+      if (kw < 0 || kw > 2) continue;
+      int64_t o = (((int64_t)n * OH + oh) * OW + ow) * C + c;
+      if (idx[o] == (uint8_t)(kh * 3 + kw)) acc += (float)gout[o];
+    }
+  }
+  gin[tid] = (T)acc;
+}
+
 }  // namespace
 
 // ------------------------------------------------------------ wrappers
@@ -207,9 +285,50 @@ void register_host_memory(at::Tensor t) {
   TORCH_CHECK(err == hipSuccess, "hipHostRegister failed: ", hipGetErrorString(err));
 }
 
+std::vector<at::Tensor> maxpool3x3s2_fwd(at::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, "maxpool: 4D CUDA tensor expected");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "maxpool: channels_last expected");
+  int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int OH = (H + 1) / 2, OW = (W + 1) / 2;
+  auto out = at::empty({N, C, OH, OW}, x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto idx = at::empty({N, C, OH, OW},
+                       x.options().dtype(at::kByte).memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * OH * OW * C;
+  int threads = 256;
+  int64_t blocks = (total + threads - 1) / threads;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, x.scalar_type(), "maxpool3x3s2_fwd", [&] {
+        hipLaunchKernelGGL(maxpool3x3s2_fwd_kernel<scalar_t>, dim3(blocks), dim3(threads), 0,
+                           stream, x.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),
+                           idx.data_ptr<uint8_t>(), N, H, W, C, OH, OW);
+      });
+  return {out, idx};
+}
+
+at::Tensor maxpool3x3s2_bwd(at::Tensor gout, at::Tensor idx, int64_t H, int64_t W) {
+  TORCH_CHECK(gout.is_cuda() && gout.dim() == 4, "maxpool bwd: 4D CUDA tensor expected");
+  auto g = gout.contiguous(at::MemoryFormat::ChannelsLast);
+  int N = g.size(0), C = g.size(1), OH = g.size(2), OW = g.size(3);
+  auto gin = at::empty({N, C, H, W}, g.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * H * W * C;
+  int threads = 256;
+  int64_t blocks = (total + threads - 1) / threads;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, g.scalar_type(), "maxpool3x3s2_bwd", [&] {
+        hipLaunchKernelGGL(maxpool3x3s2_bwd_kernel<scalar_t>, dim3(blocks), dim3(threads), 0,
+                           stream, g.data_ptr<scalar_t>(), idx.data_ptr<uint8_t>(),
+                           gin.data_ptr<scalar_t>(), N, (int)H, (int)W, C, OH, OW);
+      });
+  return gin;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "moolib_amd gfx950 HIP kernels";
   m.def("register_host_memory", &register_host_memory);
+  m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd, "NHWC 3x3/2 maxpool forward (gfx950)");
+  m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd, "NHWC 3x3/2 maxpool backward (gather, no atomics)");
   m.def("vtrace_from_log_rhos", &vtrace_from_log_rhos, "fused V-trace scan (gfx950)");
   m.def("impala_loss", &impala_loss, "fused IMPALA loss fwd+grad (gfx950)");
 }

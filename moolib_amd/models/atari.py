@@ -33,12 +33,14 @@ class ConvSection(nn.Module):
     def __init__(self, in_ch, out_ch):
         super().__init__()
         self.conv = nn.Conv2d(in_ch, out_ch, 3, stride=1, padding=1)
-        self.pool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.pool = nn.MaxPool2d(3, stride=2, padding=1)  # CPU fallback
         self.res0 = ResidualBlock(out_ch)
         self.res1 = ResidualBlock(out_ch)
 
     def forward(self, x):
-        x = self.pool(self.conv(x))
+        from moolib_amd.ops.pool import maxpool3x3s2
+
+        x = maxpool3x3s2(self.conv(x))
         x = self.res0(x)
         x = self.res1(x)
         return x

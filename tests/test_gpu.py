@@ -150,3 +150,42 @@ class TestDevicePaths:
                 opt_steps += 1
         assert opt_steps >= 3
         assert peer.stats["env_train_steps"].result() >= 3 * 5 * 8
+
+
+@gpu
+@requires_gpu
+class TestMaxPoolKernel:
+    def test_matches_torch(self):
+        import torch.nn.functional as F
+
+        from moolib_amd.ops.pool import maxpool3x3s2
+
+        torch.manual_seed(3)
+        for shape in [(2, 16, 84, 84), (4, 32, 42, 42), (672, 32, 21, 21), (1, 8, 11, 11)]:
+            x = torch.randn(*shape, device="cuda", dtype=torch.float32).to(
+                memory_format=torch.channels_last
+            ).requires_grad_()
+            x_ref = x.detach().clone().requires_grad_()
+            y = maxpool3x3s2(x)
+            y_ref = F.max_pool2d(x_ref, 3, stride=2, padding=1)
+            assert torch.equal(y, y_ref), shape
+            g = torch.randn_like(y)
+            y.backward(g)
+            y_ref.backward(g)
+            assert torch.allclose(x.grad, x_ref.grad, atol=1e-5), shape
+
+    def test_bf16(self):
+        import torch.nn.functional as F
+
+        from moolib_amd.ops.pool import maxpool3x3s2
+
+        x = torch.randn(8, 16, 84, 84, device="cuda", dtype=torch.bfloat16).to(
+            memory_format=torch.channels_last
+        ).requires_grad_()
+        x_ref = x.detach().clone().requires_grad_()
+        y = maxpool3x3s2(x)
+        y_ref = F.max_pool2d(x_ref, 3, stride=2, padding=1)
+        assert torch.equal(y, y_ref)
+        y.sum().backward()
+        y_ref.sum().backward()
+        assert torch.allclose(x.grad.float(), x_ref.grad.float())
