@@ -1,0 +1,46 @@
+"""Zero-copy same-node GPU tensor RPC via hipIpc (dmabuf) handles.
+
+The reference refuses CUDA tensors on the wire outright (rpc.cc:661-667
+`fatal`); moolib_amd stages them through the CPU by default, and this
+module provides the zero-copy path BASELINE.json names: wrap a device
+tensor in `share()` and pass it through any RPC call — the receiver
+(same node, any process) materializes a tensor aliasing the sender's HBM
+through a hipIpc memory handle, no bytes copied.
+
+Built on torch's CUDA-IPC storage sharing (torch.multiprocessing
+reductions == hipIpcGetMemHandle/hipIpcOpenMemHandle + a shared refcount
+file on ROCm; requires HSA_ENABLE_IPC_MODE_LEGACY=0 with the dmabuf
+driver).
+
+    # producer (replay server on GPU 0)
+    rpc.define("get", lambda: ipc.share(hbm_resident_batch))
+    # consumer (learner on GPU 1, same node)
+    batch = client.sync("server", "get")   # torch.Tensor aliasing GPU 0 HBM
+
+Lifetime: the producer's storage stays alive until every consumer drops
+its alias (torch's IPC refcounting). Do not send handles across nodes.
+"""
+import torch
+from torch.multiprocessing import reductions
+
+
+class SharedCudaTensor:
+    """Pickles into a hipIpc handle; unpickles into the aliasing tensor."""
+
+    def __init__(self, tensor):
+        if not tensor.is_cuda:
+            raise ValueError("share() is for device tensors; CPU tensors ship as bytes anyway")
+        self._tensor = tensor
+
+    def __reduce__(self):
+        func, args = reductions.reduce_tensor(self._tensor)
+        return (func, args)
+
+    def tensor(self):
+        return self._tensor
+
+
+def share(tensor):
+    """Wrap a CUDA tensor so RPC serialization sends a hipIpc handle
+    instead of the bytes. The result of deserialization IS the tensor."""
+    return SharedCudaTensor(tensor)
