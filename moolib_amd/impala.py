@@ -344,7 +344,10 @@ class ImpalaPeer:
         if self.scheduler is not None:
             self.scheduler.step()
         self.model_version += 1
-        self.stats["unclipped_grad_norm"] += norm.item()
+        # .item() is a device sync: sample the norm stat instead of paying
+        # the sync every step.
+        if not self.is_cuda or self.model_version % 16 == 0:
+            self.stats["unclipped_grad_norm"] += norm.item()
         self.stats["optimizer_steps"] += 1
         self.stats["model_version"] += 1
 
