@@ -16,6 +16,7 @@
 //     and writes the combined logits gradient directly (saves the autograd
 //     graph + ~15 separate elementwise/softmax kernels of the eager path).
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 
@@ -221,6 +222,29 @@ __global__ void maxpool3x3s2_bwd_kernel(const T* __restrict__ gout,
   gin[tid] = (T)acc;
 }
 
+// -------------------------------------------- frame preprocessing (fused)
+//
+// uint8 [N,C,H,W] frames -> bf16 NHWC scaled by 1/255, in one pass. The
+// eager path (to(dtype) + mul_ + contiguous(channels_last)) is 3 kernels
+// over the largest tensor in the network.
+
+__global__ void frames_u8_to_bf16_nhwc_kernel(const uint8_t* __restrict__ in,
+                                              hip_bfloat16* __restrict__ out, float scale,
+                                              int N, int C, int H, int W) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)N * C * H * W;
+  if (tid >= total) return;
+  // tid enumerates the OUTPUT (NHWC) layout for coalesced writes.
+  int c = tid % C;
+  int64_t t = tid / C;
+  int w = t % W;
+  t /= W;
+  int h = t % H;
+  int n = t / H;
+  uint8_t v = in[(((int64_t)n * C + c) * H + h) * W + w];
+  out[tid] = (hip_bfloat16)((float)v * scale);
+}
+
 }  // namespace
 
 // ------------------------------------------------------------ wrappers
@@ -285,6 +309,23 @@ void register_host_memory(at::Tensor t) {
   TORCH_CHECK(err == hipSuccess, "hipHostRegister failed: ", hipGetErrorString(err));
 }
 
+at::Tensor frames_u8_to_bf16_nhwc(at::Tensor x, double scale) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.dtype() == at::kByte,
+              "frames: 4D uint8 CUDA tensor expected");
+  auto xc = x.contiguous();
+  int N = xc.size(0), C = xc.size(1), H = xc.size(2), W = xc.size(3);
+  auto out = at::empty({N, C, H, W},
+                       xc.options().dtype(at::kBFloat16).memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * C * H * W;
+  int threads = 256;
+  int64_t blocks = (total + threads - 1) / threads;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(frames_u8_to_bf16_nhwc_kernel, dim3(blocks), dim3(threads), 0, stream,
+                     xc.data_ptr<uint8_t>(),
+                     reinterpret_cast<hip_bfloat16*>(out.data_ptr()), (float)scale, N, C, H, W);
+  return out;
+}
+
 std::vector<at::Tensor> maxpool3x3s2_fwd(at::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4, "maxpool: 4D CUDA tensor expected");
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "maxpool: channels_last expected");
@@ -329,6 +370,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("register_host_memory", &register_host_memory);
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd, "NHWC 3x3/2 maxpool forward (gfx950)");
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd, "NHWC 3x3/2 maxpool backward (gather, no atomics)");
+  m.def("frames_u8_to_bf16_nhwc", &frames_u8_to_bf16_nhwc, "fused uint8->bf16 NHWC scale");
   m.def("vtrace_from_log_rhos", &vtrace_from_log_rhos, "fused V-trace scan (gfx950)");
   m.def("impala_loss", &impala_loss, "fused IMPALA loss fwd+grad (gfx950)");
 }

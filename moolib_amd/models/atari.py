@@ -80,13 +80,28 @@ class AtariNet(nn.Module):
         x = inputs["state"]
         reward = inputs["reward"]
         T, B = x.shape[:2]
-        x = x.flatten(0, 1).to(self.fc.weight.dtype).mul_(1.0 / 255.0)
-        if x.is_cuda:
-            # NHWC: MIOpen's bf16 igemm kernels are NHWC-native; NCHW input
-            # inserts a batched_transpose around every conv. (Note: the FC
-            # input ordering then differs from the CPU/NCHW path — a
-            # self-consistent permutation of learned features.)
-            x = x.contiguous(memory_format=torch.channels_last)
+        x = x.flatten(0, 1)
+        fused = None
+        if x.is_cuda and x.dtype == torch.uint8 and (
+            torch.is_autocast_enabled() or self.fc.weight.dtype == torch.bfloat16
+        ):
+            try:
+                from moolib_amd import _kernels
+
+                fused = _kernels.frames_u8_to_bf16_nhwc(x, 1.0 / 255.0)
+            except ImportError:
+                fused = None
+        if fused is not None:
+            x = fused
+        else:
+            x = x.to(self.fc.weight.dtype).mul_(1.0 / 255.0)
+            if x.is_cuda:
+                # NHWC: MIOpen's bf16 igemm kernels are NHWC-native; NCHW
+                # input inserts a batched_transpose around every conv.
+                # (Note: the FC input ordering then differs from the
+                # CPU/NCHW path — a self-consistent permutation of learned
+                # features.)
+                x = x.contiguous(memory_format=torch.channels_last)
 
         for s in self.sections:
             x = s(x)

@@ -189,3 +189,17 @@ class TestMaxPoolKernel:
         y.sum().backward()
         y_ref.sum().backward()
         assert torch.allclose(x.grad.float(), x_ref.grad.float())
+
+
+@gpu
+@requires_gpu
+class TestFramesKernel:
+    def test_matches_eager(self):
+        from moolib_amd import _kernels
+
+        x = torch.randint(0, 256, (7, 4, 84, 84), dtype=torch.uint8, device="cuda")
+        got = _kernels.frames_u8_to_bf16_nhwc(x, 1.0 / 255.0)
+        want = (x.float() / 255.0).bfloat16().contiguous(memory_format=torch.channels_last)
+        assert got.dtype == torch.bfloat16
+        assert got.is_contiguous(memory_format=torch.channels_last)
+        assert torch.allclose(got.float(), want.float(), atol=1 / 255.0)
