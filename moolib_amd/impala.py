@@ -17,6 +17,7 @@ MI355X design:
 """
 import contextlib
 import dataclasses
+import os
 import time
 from typing import Any, Optional
 
@@ -185,6 +186,11 @@ class ImpalaPeer:
         self.next_env_index = 0
         self.model_version = 0
         self.is_cuda = torch.device(cfg.device).type == "cuda"
+        # env kill-switches for debugging/bisection
+        if os.environ.get("MOOLIB_AMD_NO_GRAPHS"):
+            cfg.graph_actor = cfg.graph_learner = False
+        if os.environ.get("MOOLIB_AMD_NO_SIDE_STREAM"):
+            cfg.actor_side_stream = False
         self.autocast = cfg.autocast_bf16 and self.is_cuda
         if self.is_cuda and cfg.channels_last:
             self.model.to(memory_format=torch.channels_last)

@@ -82,7 +82,11 @@ class AtariNet(nn.Module):
         T, B = x.shape[:2]
         x = x.flatten(0, 1)
         fused = None
-        if x.is_cuda and x.dtype == torch.uint8 and (
+        import os as _os
+
+        if x.is_cuda and x.dtype == torch.uint8 and not _os.environ.get(
+            "MOOLIB_AMD_NO_FRAMES_KERNEL"
+        ) and (
             torch.is_autocast_enabled() or self.fc.weight.dtype == torch.bfloat16
         ):
             try:

@@ -6,6 +6,8 @@ a per-output window index in the forward and gathers in the backward (each
 input element checks its <=4 covering windows) — atomic-free and
 bandwidth-bound. Falls back to F.max_pool2d off-GPU.
 """
+import os
+
 import torch
 import torch.nn.functional as F
 
@@ -37,6 +39,6 @@ class _MaxPool3x3s2(torch.autograd.Function):
 
 def maxpool3x3s2(x):
     """max_pool2d(x, 3, stride=2, padding=1) — fused NHWC path on MI355X."""
-    if x.is_cuda and _kernels() is not None:
+    if x.is_cuda and _kernels() is not None and not os.environ.get("MOOLIB_AMD_NO_POOL_KERNEL"):
         return _MaxPool3x3s2.apply(x)
     return F.max_pool2d(x, 3, stride=2, padding=1)
