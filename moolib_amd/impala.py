@@ -62,7 +62,9 @@ class ImpalaConfig:
     graph_learner: bool = True   # hipGraph-capture the learner fwd+bwd
     pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
     shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
-    actor_side_stream: bool = True   # overlap actor H2D+forward with the learner stream
+    actor_side_stream: bool = False  # overlap actor work on a side stream. Off by default:
+    #   concurrent hipGraph replays on two streams intermittently fault on ROCm 7.0
+    #   (HSA_STATUS_ERROR_EXCEPTION, ~25%% of runs); measured no throughput win.
     max_learn_backlog: int = 4       # stop acting when this many learn batches are queued
     group_timeout: float = 10.0      # broker eviction / allreduce timeout (seconds)
     bf16_weights: bool = False       # forward on a bf16 shadow of the fp32 master weights (measured neutral vs autocast)
