@@ -222,6 +222,14 @@ class ImpalaPeer:
             torch.cuda.Stream() if (self.is_cuda and cfg.actor_side_stream) else None
         )
         self._batch_events = []
+        if self.is_cuda and hasattr(self.fwd_model, "sample_generator"):
+            self._actor_rng = torch.Generator(device=cfg.device)
+            self._actor_rng.manual_seed(int(torch.seed()) % (2**62))
+            self._learn_rng = torch.Generator(device=cfg.device)
+            self._learn_rng.manual_seed(int(torch.seed()) % (2**62))
+        else:
+            self._actor_rng = None
+            self._learn_rng = None
         from moolib_amd.parallel.graphs import GraphedCall
 
         self._actor_call = (
@@ -273,6 +281,8 @@ class ImpalaPeer:
         fixed shapes, writes gradients into the (stable) param.grad tensors."""
         cfg = self.cfg
         model = self.fwd_model
+        if self._learn_rng is not None:
+            model.sample_generator = self._learn_rng
         if self.bf16_shadow:
             for pb in self._fwd_params:
                 if pb.grad is not None:
@@ -336,6 +346,8 @@ class ImpalaPeer:
     def _actor_fn(self, inputs):
         """No-grad actor forward on [1, B]. hipGraph-capturable."""
         env_outputs, core_state = inputs["env"], inputs["core"]
+        if self._actor_rng is not None:
+            self.fwd_model.sample_generator = self._actor_rng
         with torch.no_grad(), torch.autocast(
             "cuda", dtype=torch.bfloat16, enabled=self.autocast, cache_enabled=False
         ):

@@ -67,6 +67,11 @@ class AtariNet(nn.Module):
             core_in = 256
         self.policy = nn.Linear(core_in, num_actions)
         self.baseline = nn.Linear(core_in, 1)
+        # Optional dedicated RNG for the in-forward multinomial: hipGraph
+        # replays of two captures racing on the DEFAULT generator's device
+        # state is what forced the actor side stream off; per-capture
+        # generators make concurrent replays sound.
+        self.sample_generator = None
 
     def initial_state(self, batch_size=1):
         if not self.use_lstm:
@@ -135,7 +140,9 @@ class AtariNet(nn.Module):
 
         policy_logits = self.policy(core_output).float()
         baseline = self.baseline(core_output).float()
-        action = torch.multinomial(F.softmax(policy_logits, dim=-1), num_samples=1)
+        action = torch.multinomial(
+            F.softmax(policy_logits, dim=-1), num_samples=1, generator=self.sample_generator
+        )
 
         out = dict(
             policy_logits=policy_logits.view(T, B, self.num_actions),
