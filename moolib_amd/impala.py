@@ -193,6 +193,8 @@ class ImpalaPeer:
             cfg.graph_actor = cfg.graph_learner = False
         if os.environ.get("MOOLIB_AMD_NO_SIDE_STREAM"):
             cfg.actor_side_stream = False
+        if os.environ.get("MOOLIB_AMD_FORCE_SIDE_STREAM"):
+            cfg.actor_side_stream = True
         self.autocast = cfg.autocast_bf16 and self.is_cuda
         if self.is_cuda and cfg.channels_last:
             self.model.to(memory_format=torch.channels_last)
