@@ -247,7 +247,57 @@ __global__ void frames_u8_to_bf16_nhwc_kernel(const uint8_t* __restrict__ in,
 
 }  // namespace
 
+#include "lstm.hip.inc"
+
 // ------------------------------------------------------------ wrappers
+
+std::vector<at::Tensor> lstm_fused_fwd(at::Tensor X, at::Tensor notdone, at::Tensor h0,
+                                       at::Tensor c0, at::Tensor whhPacked) {
+  TORCH_CHECK(X.is_cuda() && X.dtype() == at::kBFloat16 && X.dim() == 3, "lstm fwd: X [T,B,4H] bf16");
+  int T = X.size(0), B = X.size(1);
+  TORCH_CHECK(X.size(2) == lstm::kG, "lstm fwd: hidden size must be 256");
+  auto opts = X.options();
+  auto fopts = opts.dtype(at::kFloat);
+  auto H = at::empty({T, B, lstm::kH}, opts);
+  auto gates = at::empty({T, B, lstm::kG}, opts);
+  auto cs = at::empty({T, B, lstm::kH}, fopts);
+  auto hT = at::empty({B, lstm::kH}, opts);
+  auto cT = at::empty({B, lstm::kH}, fopts);
+  int blocks = (B + lstm::kBM - 1) / lstm::kBM;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(lstm::lstm_fwd_kernel, dim3(blocks), dim3(lstm::kThreads), 0, stream,
+                     reinterpret_cast<const lstm::bf16*>(X.data_ptr()),
+                     notdone.data_ptr<float>(),
+                     reinterpret_cast<const lstm::bf16*>(h0.data_ptr()), c0.data_ptr<float>(),
+                     reinterpret_cast<const lstm::bf16*>(whhPacked.data_ptr()),
+                     reinterpret_cast<lstm::bf16*>(H.data_ptr()),
+                     reinterpret_cast<lstm::bf16*>(gates.data_ptr()), cs.data_ptr<float>(),
+                     reinterpret_cast<lstm::bf16*>(hT.data_ptr()), cT.data_ptr<float>(), T, B);
+  return {H, gates, cs, hT, cT};
+}
+
+std::vector<at::Tensor> lstm_fused_bwd(at::Tensor gates, at::Tensor cs, at::Tensor c0,
+                                       at::Tensor notdone, at::Tensor dH, at::Tensor dhT,
+                                       at::Tensor dcT, at::Tensor whhPackedN) {
+  int T = gates.size(0), B = gates.size(1);
+  auto opts = gates.options();
+  auto fopts = opts.dtype(at::kFloat);
+  auto dG = at::empty({T, B, lstm::kG}, opts);
+  auto dh0 = at::empty({B, lstm::kH}, opts);
+  auto dc0 = at::empty({B, lstm::kH}, fopts);
+  int blocks = (B + lstm::kBM - 1) / lstm::kBM;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(lstm::lstm_bwd_kernel, dim3(blocks), dim3(lstm::kThreads), 0, stream,
+                     reinterpret_cast<const lstm::bf16*>(gates.data_ptr()), cs.data_ptr<float>(),
+                     c0.data_ptr<float>(), notdone.data_ptr<float>(),
+                     reinterpret_cast<const lstm::bf16*>(dH.data_ptr()),
+                     dhT.defined() ? reinterpret_cast<const lstm::bf16*>(dhT.data_ptr()) : nullptr,
+                     dcT.defined() ? dcT.data_ptr<float>() : nullptr,
+                     reinterpret_cast<const lstm::bf16*>(whhPackedN.data_ptr()),
+                     reinterpret_cast<lstm::bf16*>(dG.data_ptr()),
+                     reinterpret_cast<lstm::bf16*>(dh0.data_ptr()), dc0.data_ptr<float>(), T, B);
+  return {dG, dh0, dc0};
+}
 
 std::vector<at::Tensor> vtrace_from_log_rhos(at::Tensor log_rhos, at::Tensor discounts,
                                              at::Tensor rewards, at::Tensor values,
@@ -371,6 +421,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd, "NHWC 3x3/2 maxpool forward (gfx950)");
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd, "NHWC 3x3/2 maxpool backward (gather, no atomics)");
   m.def("frames_u8_to_bf16_nhwc", &frames_u8_to_bf16_nhwc, "fused uint8->bf16 NHWC scale");
+  m.def("lstm_fused_fwd", &lstm_fused_fwd, "fused masked LSTM sequence scan fwd (MFMA, gfx950)");
+  m.def("lstm_fused_bwd", &lstm_fused_bwd, "fused masked LSTM sequence scan bwd (MFMA, gfx950)");
   m.def("vtrace_from_log_rhos", &vtrace_from_log_rhos, "fused V-trace scan (gfx950)");
   m.def("impala_loss", &impala_loss, "fused IMPALA loss fwd+grad (gfx950)");
 }

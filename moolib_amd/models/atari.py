@@ -127,14 +127,38 @@ class AtariNet(nn.Module):
         if self.use_lstm:
             done = inputs["done"]
             core_input = core_input.view(T, B, -1)
-            notdone = (~done).to(core_input.dtype)
-            outputs = []
-            for t in range(T):
-                nd = notdone[t].view(1, -1, 1)
-                core_state = tuple(nd * s for s in core_state)
-                out_t, core_state = self.core(core_input[t : t + 1], core_state)
-                outputs.append(out_t)
-            core_output = torch.cat(outputs).flatten(0, 1)
+            from moolib_amd.ops import lstm as lstm_ops
+
+            if lstm_ops.available(self.core.hidden_size, core_input.device) and (
+                torch.is_autocast_enabled() or core_input.dtype == torch.bfloat16
+            ):
+                # fused MFMA sequence scan: one GEMM for x@W_ih over all T,
+                # one kernel for the masked recurrence (fwd and bwd each).
+                Xg = F.linear(
+                    core_input.reshape(T * B, -1),
+                    self.core.weight_ih_l0,
+                    self.core.bias_ih_l0 + self.core.bias_hh_l0,
+                ).view(T, B, -1).to(torch.bfloat16)
+                notdone_f = (~done).float()
+                h0 = core_state[0][0].to(torch.bfloat16)
+                c0 = core_state[1][0].float()
+                H, hT, cT = lstm_ops.fused_lstm_scan(
+                    Xg, notdone_f, h0, c0, self.core.weight_hh_l0
+                )
+                core_output = H.flatten(0, 1)
+                core_state = (
+                    hT.unsqueeze(0).to(core_state[0].dtype),
+                    cT.unsqueeze(0).to(core_state[1].dtype),
+                )
+            else:
+                notdone = (~done).to(core_input.dtype)
+                outputs = []
+                for t in range(T):
+                    nd = notdone[t].view(1, -1, 1)
+                    core_state = tuple(nd * s for s in core_state)
+                    out_t, core_state = self.core(core_input[t : t + 1], core_state)
+                    outputs.append(out_t)
+                core_output = torch.cat(outputs).flatten(0, 1)
         else:
             core_output = core_input
 
