@@ -237,12 +237,22 @@ class ImpalaPeer:
         from moolib_amd.parallel.graphs import GraphedCall
 
         self._actor_call = (
-            GraphedCall(self._actor_fn, warmup=3, name="actor_fwd")
+            GraphedCall(
+                self._actor_fn,
+                warmup=3,
+                name="actor_fwd",
+                generators=[self._actor_rng] if self._actor_rng is not None else (),
+            )
             if (self.is_cuda and cfg.graph_actor)
             else self._actor_fn
         )
         self._learn_call = (
-            GraphedCall(self._learn_fn, warmup=3, name="learner_fwd_bwd")
+            GraphedCall(
+                self._learn_fn,
+                warmup=3,
+                name="learner_fwd_bwd",
+                generators=[self._learn_rng] if self._learn_rng is not None else (),
+            )
             if (self.is_cuda and cfg.graph_learner)
             else self._learn_fn
         )

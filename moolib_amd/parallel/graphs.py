@@ -28,10 +28,11 @@ class GraphedCall:
     before the next call. Falls back to eager permanently if capture fails.
     """
 
-    def __init__(self, fn, warmup=3, name="graph"):
+    def __init__(self, fn, warmup=3, name="graph", generators=()):
         self.fn = fn
         self.warmup = warmup
         self.name = name
+        self.generators = list(generators)  # non-default RNGs used inside fn
         self.calls = 0
         self.graph = None
         self.static_in = None
@@ -49,6 +50,10 @@ class GraphedCall:
                 )
                 torch.cuda.synchronize()
                 g = torch.cuda.CUDAGraph()
+                for gen in self.generators:
+                    # dedicated (non-default) generators must be registered
+                    # before capture so replays advance their offsets
+                    g.register_generator_state(gen.graphsafe_get_state())
                 with torch.cuda.graph(g):
                     self.static_out = self.fn(self.static_in)
                 self.graph = g
