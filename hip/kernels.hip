@@ -152,22 +152,35 @@ __global__ void impala_loss_kernel(const float* __restrict__ logits,
 // backward (each input element checks its <=4 covering windows) — no
 // atomics, bandwidth-bound.
 
+// Vectorized over 8 NHWC channels per thread: all loads/stores are 16-byte
+// (bf16x8) or 8-byte (idx) contiguous chunks.
+template <typename T>
+struct Vec8 {
+  T v[8];
+};
+
 template <typename T>
 __global__ void maxpool3x3s2_fwd_kernel(const T* __restrict__ in, T* __restrict__ out,
                                         uint8_t* __restrict__ idx, int N, int H, int W, int C,
                                         int OH, int OW) {
+  int C8 = C >> 3;
   int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t total = (int64_t)N * OH * OW * C;
+  int64_t total = (int64_t)N * OH * OW * C8;
   if (tid >= total) return;
-  int c = tid % C;
-  int64_t t = tid / C;
+  int c8 = tid % C8;
+  int64_t t = tid / C8;
   int ow = t % OW;
   t /= OW;
   int oh = t % OH;
   int n = t / OH;
 
-  float best = -INFINITY;
-  int besti = 0;
+  float best[8];
+  int besti[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    best[j] = -INFINITY;
+    besti[j] = 0;
+  }
 #pragma unroll
   for (int kh = 0; kh < 3; ++kh) {
     int ih = oh * 2 - 1 + kh;
@@ -176,38 +189,50 @@ __global__ void maxpool3x3s2_fwd_kernel(const T* __restrict__ in, T* __restrict_
     for (int kw = 0; kw < 3; ++kw) {
       int iw = ow * 2 - 1 + kw;
       if (iw < 0 || iw >= W) continue;
-      float v = (float)in[(((int64_t)n * H + ih) * W + iw) * C + c];
-      if (v > best) {
-        best = v;
-        besti = kh * 3 + kw;
+      const Vec8<T> vin =
+          *reinterpret_cast<const Vec8<T>*>(in + ((((int64_t)n * H + ih) * W + iw) * C8 + c8) * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = (float)vin.v[j];
+        if (v > best[j]) {
+          best[j] = v;
+          besti[j] = kh * 3 + kw;
+        }
       }
     }
   }
-  out[tid] = (T)best;
-  idx[tid] = (uint8_t)besti;
+  Vec8<T> vout;
+  Vec8<uint8_t> vidx;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    vout.v[j] = (T)best[j];
+    vidx.v[j] = (uint8_t)besti[j];
+  }
+  *reinterpret_cast<Vec8<T>*>(out + tid * 8) = vout;
+  *reinterpret_cast<Vec8<uint8_t>*>(idx + tid * 8) = vidx;
 }
 
 template <typename T>
 __global__ void maxpool3x3s2_bwd_kernel(const T* __restrict__ gout,
                                         const uint8_t* __restrict__ idx, T* __restrict__ gin,
                                         int N, int H, int W, int C, int OH, int OW) {
+  int C8 = C >> 3;
   int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t total = (int64_t)N * H * W * C;
+  int64_t total = (int64_t)N * H * W * C8;
   if (tid >= total) return;
-  int c = tid % C;
-  int64_t t = tid / C;
+  int c8 = tid % C8;
+  int64_t t = tid / C8;
   int iw = t % W;
   t /= W;
   int ih = t % H;
   int n = t / H;
 
-  float acc = 0.f;
-  // windows (oh, ow) covering (ih, iw): oh*2-1 <= ih <= oh*2+1
-  int oh_lo = (ih - 1 + 1) / 2;  // ceil((ih-1)/2) for ih>=1; 0 for ih==0
-  if (ih == 0) oh_lo = 0;
+  float acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  int oh_lo = (ih == 0) ? 0 : ih / 2;  // ceil((ih-1)/2)
   int oh_hi = (ih + 1) / 2;
-  int ow_lo = (iw == 0) ? 0 : (iw - 1 + 1) / 2;
-  if (iw == 0) ow_lo = 0;
+  int ow_lo = (iw == 0) ? 0 : iw / 2;
   int ow_hi = (iw + 1) / 2;
   for (int oh = oh_lo; oh <= oh_hi && oh < OH; ++oh) {
     int kh = ih - (oh * 2 - 1);
@@ -215,11 +240,20 @@ __global__ void maxpool3x3s2_bwd_kernel(const T* __restrict__ gout,
     for (int ow = ow_lo; ow <= ow_hi && ow < OW; ++ow) {
       int kw = iw - (ow * 2 - 1);
       if (kw < 0 || kw > 2) continue;
-      int64_t o = (((int64_t)n * OH + oh) * OW + ow) * C + c;
-      if (idx[o] == (uint8_t)(kh * 3 + kw)) acc += (float)gout[o];
+      int64_t o = (((int64_t)n * OH + oh) * OW + ow) * C8 + c8;
+      const Vec8<uint8_t> vidx = *reinterpret_cast<const Vec8<uint8_t>*>(idx + o * 8);
+      const Vec8<T> vg = *reinterpret_cast<const Vec8<T>*>(gout + o * 8);
+      uint8_t code = (uint8_t)(kh * 3 + kw);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        if (vidx.v[j] == code) acc[j] += (float)vg.v[j];
+      }
     }
   }
-  gin[tid] = (T)acc;
+  Vec8<T> vout;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) vout.v[j] = (T)acc[j];
+  *reinterpret_cast<Vec8<T>*>(gin + tid * 8) = vout;
 }
 
 // -------------------------------------------- frame preprocessing (fused)
@@ -384,7 +418,8 @@ std::vector<at::Tensor> maxpool3x3s2_fwd(at::Tensor x) {
   auto out = at::empty({N, C, OH, OW}, x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto idx = at::empty({N, C, OH, OW},
                        x.options().dtype(at::kByte).memory_format(at::MemoryFormat::ChannelsLast));
-  int64_t total = (int64_t)N * OH * OW * C;
+  TORCH_CHECK(C % 8 == 0, "maxpool: channels must be a multiple of 8");
+  int64_t total = (int64_t)N * OH * OW * (C / 8);
   int threads = 256;
   int64_t blocks = (total + threads - 1) / threads;
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
@@ -402,7 +437,8 @@ at::Tensor maxpool3x3s2_bwd(at::Tensor gout, at::Tensor idx, int64_t H, int64_t 
   auto g = gout.contiguous(at::MemoryFormat::ChannelsLast);
   int N = g.size(0), C = g.size(1), OH = g.size(2), OW = g.size(3);
   auto gin = at::empty({N, C, H, W}, g.options().memory_format(at::MemoryFormat::ChannelsLast));
-  int64_t total = (int64_t)N * H * W * C;
+  TORCH_CHECK(C % 8 == 0, "maxpool bwd: channels must be a multiple of 8");
+  int64_t total = (int64_t)N * H * W * (C / 8);
   int threads = 256;
   int64_t blocks = (total + threads - 1) / threads;
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();

@@ -39,6 +39,11 @@ class _MaxPool3x3s2(torch.autograd.Function):
 
 def maxpool3x3s2(x):
     """max_pool2d(x, 3, stride=2, padding=1) — fused NHWC path on MI355X."""
-    if x.is_cuda and _kernels() is not None and not os.environ.get("MOOLIB_AMD_NO_POOL_KERNEL"):
+    if (
+        x.is_cuda
+        and x.shape[1] % 8 == 0
+        and _kernels() is not None
+        and not os.environ.get("MOOLIB_AMD_NO_POOL_KERNEL")
+    ):
         return _MaxPool3x3s2.apply(x)
     return F.max_pool2d(x, 3, stride=2, padding=1)
