@@ -62,11 +62,12 @@ class ImpalaConfig:
     graph_learner: bool = True   # hipGraph-capture the learner fwd+bwd
     pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
     shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
-    actor_side_stream: bool = True   # overlap actor H2D+forward with the learner stream
-    #   (requires the per-capture RNG generators below: concurrent hipGraph
-    #   replays racing on the DEFAULT generator's device state caused
-    #   intermittent HSA memory faults; with dedicated generators 10/10
-    #   stress runs pass and the overlap is worth ~7%)
+    actor_side_stream: bool = False  # overlap actor work on a side HIP stream.
+    #   OFF by default: concurrent hipGraph replays on two streams still hit
+    #   an intermittent HSA memory fault (~1 in 8 runs) on ROCm 7.0 even
+    #   with per-capture RNG generators registered; single-stream replay is
+    #   9/9 stable and costs only ~2%. Re-enable with
+    #   MOOLIB_AMD_FORCE_SIDE_STREAM=1 to investigate (round 2).
     max_learn_backlog: int = 4       # stop acting when this many learn batches are queued
     group_timeout: float = 10.0      # broker eviction / allreduce timeout (seconds)
     bf16_weights: bool = True        # forward on bf16 shadow weights (fp32 master for optimizer/sync):
