@@ -228,3 +228,29 @@ class TestBatchedDefine:
             msg, s = f.result()
             assert msg == "m"
             assert torch.allclose(s, tensors[i].sum())
+
+
+class TestDynamicBatching:
+    def test_partial_batch_flushes(self):
+        """dynamic_batching=True must flush partial batches after max_latency."""
+        host = moolib_amd.Rpc()
+        host.set_name("host")
+        addr = host.listen("127.0.0.1:0")[0]
+        seen_sizes = []
+
+        def f(t):
+            seen_sizes.append(t.shape[0])
+            return t.sum(1)
+
+        host.define("f", f, batch_size=8, dynamic_batching=True)
+        clients = []
+        futures = []
+        for i in range(3):  # fewer than batch_size
+            c = moolib_amd.Rpc()
+            c.set_timeout(15)
+            c.connect(addr)
+            clients.append(c)
+            futures.append(c.async_("host", "f", torch.ones(2)))
+        for f_ in futures:
+            assert float(f_.result()) == 2.0
+        assert seen_sizes == [3]
