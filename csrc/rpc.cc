@@ -413,12 +413,16 @@ void Rpc::sendRequest(const std::string& peerName, const std::string& funcName, 
   ensureListeningLocked();
   Outgoing& rec = outgoing_[rid];
   rec.rid = rid;
+  rec.sentAt = now();
   rec.peerName = peerName;
   rec.funcName = funcName;
   rec.frame = f;
   rec.deadline = now() + std::chrono::duration_cast<Clock::duration>(std::chrono::duration<double>(timeoutS));
   rec.cb = std::move(cb);
   PeerInfo& p = getPeer(peerName);
+  uint64_t fbytes = f.payload.size();
+  for (auto& t : f.tensors) fbytes += t.nbytes();
+  p.bytesSent += fbytes;
   if (p.activeConn != 0) {
     rec.sentOn = p.activeConn;
     engine_->send(p.activeConn, std::move(f));
@@ -502,6 +506,13 @@ void Rpc::handleResponse(ConnId id, Frame&& f, bool isError) {
     cb = std::move(it->second.cb);
     peerName = it->second.peerName;
     funcName = it->second.funcName;
+    double lat = secondsSince(it->second.sentAt);
+    PeerInfo& p = getPeer(peerName);
+    p.latencyEma = p.latencyEma == 0 ? lat : p.latencyEma * 0.9 + lat * 0.1;
+    ++p.recvCount;
+    uint64_t fbytes = f.payload.size();
+    for (auto& t : f.tensors) fbytes += t.nbytes();
+    p.bytesRecv += fbytes;
     outgoing_.erase(it);
   }
   if (!cb) return;
@@ -647,7 +658,9 @@ std::string Rpc::debugInfo() {
   }
   os << "  peers: " << peers_.size() << "\n";
   for (auto& [name, p] : peers_) {
-    os << "    '" << name << "' conn=" << p.activeConn << " sends=" << p.sendCount << " addrs=[";
+    os << "    '" << name << "' conn=" << p.activeConn << " reqs=" << p.sendCount << "/"
+       << p.recvCount << " bytes tx/rx=" << p.bytesSent << "/" << p.bytesRecv
+       << " latency_ema=" << p.latencyEma * 1000 << "ms addrs=[";
     for (auto& a : p.addrs) os << a << ",";
     os << "]\n";
   }

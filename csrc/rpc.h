@@ -94,10 +94,15 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
     TimePoint lastConnectAttempt{};
     TimePoint lastFindBroadcast{};
     uint64_t sendCount = 0;
+    uint64_t recvCount = 0;
+    uint64_t bytesSent = 0;
+    uint64_t bytesRecv = 0;
+    double latencyEma = 0;  // seconds, over request->response round trips
   };
 
   struct Outgoing {
     uint64_t rid;
+    TimePoint sentAt{};
     std::string peerName;
     std::string funcName;
     Frame frame;  // full request for resend
