@@ -31,22 +31,26 @@ Accumulator::Accumulator(std::string name, std::vector<at::Tensor> params,
   for (auto& p : allParams_) {
     if (p.requires_grad()) params_.push_back(p);
   }
-  // One flat on-device bucket sized for every requires-grad parameter. On
-  // MI355X this is the tensor RCCL reduces over xGMI; 288 GB HBM3E makes a
-  // single resident bucket the right default even for very large models.
+  // Flat on-device buckets sized for every requires-grad parameter. On
+  // MI355X these are the tensors RCCL reduces over xGMI; 288 GB HBM3E makes
+  // resident buckets the right default even for very large models.
   int64_t total = 0;
   for (auto& p : params_) {
     offsets_.push_back(total);
     numels_.push_back(p.numel());
     total += p.numel();
   }
-  if (!params_.empty()) {
-    NoGrad ng;
-    flat_ = at::zeros({std::max<int64_t>(total, 1)},
-                      at::TensorOptions().dtype(params_[0].scalar_type()).device(params_[0].device()));
-  } else {
-    flat_ = at::zeros({1});
-  }
+  slots_.resize(1);
+  slots_[0].flat = makeFlatLocked();
+}
+
+at::Tensor Accumulator::makeFlatLocked() {
+  NoGrad ng;
+  int64_t total = offsets_.empty() ? 1 : offsets_.back() + numels_.back();
+  auto opts = params_.empty()
+                  ? at::TensorOptions().dtype(at::kFloat)
+                  : at::TensorOptions().dtype(params_[0].scalar_type()).device(params_[0].device());
+  return at::zeros({std::max<int64_t>(total, 1)}, opts);
 }
 
 Accumulator::~Accumulator() {
@@ -119,19 +123,28 @@ void Accumulator::connect(const std::string& addr) { rpc_->connect(addr); }
 void Accumulator::resetLocked(const char* why) {
   MRL_LOG_INFO("accumulator '%s': reset (%s)", name_.c_str(), why);
   epoch_++;
-  gradPhase_ = GradPhase::wantDecision;
   decided_ = false;
-  newBatch_ = newGrads_ = newSkipped_ = 0;
-  totBatch_ = totGrads_ = totSkipped_ = 0;
   hasGradients_ = false;
-  hookStartPending_ = false;
-  if (hookPoll_) {
-    // Never abandon an in-flight collective: the peers' call sequences must
-    // stay aligned. Keep polling it to completion (drain), then discard.
-    hookAbandoned_ = true;
-  } else {
-    NoGrad ng;
-    flat_.zero_();
+  results_.clear();
+  slotCursor_ = 0;
+  hookLaunchCounter_ = 0;
+  hookLaunchedUpTo_ = 0;
+  for (auto& s : slots_) {
+    s.newBatch = s.newGrads = s.newSkipped = 0;
+    s.totBatch = s.totGrads = s.totSkipped = 0;
+    s.phase = GradPhase::wantDecision;
+    s.hookPending = false;
+    if (s.hookPoll) {
+      // Never abandon an in-flight collective: peers' call sequences must
+      // stay aligned. Drain it; its buffer (the old flat) belongs to the
+      // closure now; the slot gets a fresh bucket.
+      drains_.push_back(std::move(s.hookPoll));
+      s.hookPoll = nullptr;
+      s.flat = makeFlatLocked();
+    } else {
+      NoGrad ng;
+      s.flat.zero_();
+    }
   }
   havePendingModel_ = false;
   hasNewState_ = false;
@@ -139,6 +152,13 @@ void Accumulator::resetLocked(const char* why) {
   modelRequestSent_ = false;
   leader_.clear();
   isLeader_ = false;
+}
+
+void Accumulator::failAndResyncLocked(const char* why) {
+  MRL_LOG_INFO("accumulator '%s': %s; resync", name_.c_str(), why);
+  phase_ = Phase::inactive;
+  syncSeen_ = 0;
+  group_->requestResync();
 }
 
 void Accumulator::startElectionLocked() {
@@ -156,11 +176,7 @@ void Accumulator::startElectionLocked() {
                       std::lock_guard<std::mutex> lk(self->mu_);
                       if (epoch != self->epoch_ || self->phase_ != Phase::electing) return;
                       if (err) {
-                        MRL_LOG_INFO("accumulator '%s': election failed (%s); will retry",
-                                     self->name_.c_str(), err->c_str());
-                        self->phase_ = Phase::inactive;
-                        self->syncSeen_ = 0;
-                        self->group_->requestResync();
+                        self->failAndResyncLocked(("election failed: " + *err).c_str());
                         return;
                       }
                       self->leader_ = rv->leaderName;
@@ -183,23 +199,17 @@ void Accumulator::update() {
   uint64_t sid = group_->syncId();
 
   std::unique_lock<std::mutex> lk(mu_);
-  // Drain an abandoned hook collective regardless of phase: the peers'
-  // collective call sequences must stay aligned, so it is polled to
-  // completion and its (mixed-round) result discarded.
-  if (hookPoll_ && hookAbandoned_) {
+  // Drain abandoned hook collectives regardless of phase.
+  for (auto it = drains_.begin(); it != drains_.end();) {
     bool done = true;
     try {
-      done = hookPoll_();
+      done = (*it)();
     } catch (const std::exception& e) {
       MRL_LOG_ERROR("abandoned hook drain failed: %s", e.what());
     }
-    if (done) {
-      hookPoll_ = nullptr;
-      hookAbandoned_ = false;
-      NoGrad ng;
-      flat_.zero_();
-    }
+    it = done ? drains_.erase(it) : ++it;
   }
+
   if (!active) {
     if (phase_ != Phase::inactive) resetLocked("group inactive");
     phase_ = Phase::inactive;
@@ -251,163 +261,187 @@ void Accumulator::update() {
   }
 
   if (phase_ == Phase::running) {
-    if (gradPhase_ == GradPhase::reducing &&
-        secondsSince(gradPhaseStarted_) > 120.0) {
-      MRL_LOG_ERROR("accumulator '%s': gradient reduce wedged >120s; forcing resync",
-                    name_.c_str());
-      if (hookPoll_) hookAbandoned_ = true;
-      phase_ = Phase::inactive;
-      syncSeen_ = 0;
-      group_->requestResync();
-      return;
+    // Watchdog: a wedged reduction forces a cluster-wide resync.
+    for (auto& s : slots_) {
+      if (s.phase == GradPhase::reducing && secondsSince(s.started) > 120.0) {
+        MRL_LOG_ERROR("accumulator '%s': gradient reduce wedged >120s", name_.c_str());
+        failAndResyncLocked("reduce wedged");
+        return;
+      }
     }
-    if (gradPhase_ == GradPhase::wantDecision && decided_) {
+    GradSlot& cur = slots_[slotCursor_];
+    if (cur.phase == GradPhase::wantDecision && decided_) {
       startCountRoundLocked();
-    } else if (gradPhase_ == GradPhase::reducing && hookStartPending_ && !hookPoll_) {
-      hookStartPending_ = false;
-      try {
-        hookPoll_ = hook_(flat_);
-      } catch (const std::exception& e) {
-        MRL_LOG_ERROR("local reduce hook failed: %s", e.what());
-        phase_ = Phase::inactive;
-        syncSeen_ = 0;
-        group_->requestResync();
-        return;
-      }
-    } else if (gradPhase_ == GradPhase::reducing && hookPoll_ && !hookAbandoned_) {
-      bool done = false;
-      try {
-        done = hookPoll_();
-      } catch (const std::exception& e) {
-        MRL_LOG_ERROR("local reduce hook poll failed: %s", e.what());
-        hookPoll_ = nullptr;
-        phase_ = Phase::inactive;
-        syncSeen_ = 0;
-        group_->requestResync();
-        return;
-      }
-      if (done) {
-        hookPoll_ = nullptr;
-        applyGradResultLocked(flat_);
+    }
+    tryLaunchHooksLocked();
+    // Poll launched hook collectives.
+    for (size_t si = 0; si < slots_.size(); ++si) {
+      GradSlot& s = slots_[si];
+      if (s.phase == GradPhase::reducing && s.hookPoll && !s.hookPending) {
+        bool done = false;
+        try {
+          done = s.hookPoll();
+        } catch (const std::exception& e) {
+          s.hookPoll = nullptr;
+          MRL_LOG_ERROR("local reduce hook poll failed: %s", e.what());
+          failAndResyncLocked("hook poll failed");
+          return;
+        }
+        if (done) {
+          s.hookPoll = nullptr;
+          NoGrad ng;
+          at::Tensor result = s.flat.clone();
+          completeSlotLocked(si, std::move(result));
+        }
       }
     }
+    applyPendingLocked();
     maybeSendModelUpdatesLocked();
   }
 }
 
 void Accumulator::startCountRoundLocked() {
-  gradPhase_ = GradPhase::counting;
-  gradPhaseStarted_ = now();
+  size_t si = slotCursor_;
+  GradSlot& s = slots_[si];
+  s.phase = GradPhase::counting;
+  s.started = now();
   ReduceValue v;
   v.kind = ReduceValue::counts;
-  v.batchSize = newBatch_;
-  v.numGradients = newGrads_;
-  v.numSkipped = newSkipped_;
-  newBatch_ = newGrads_ = newSkipped_ = 0;
+  v.batchSize = s.newBatch;
+  v.numGradients = s.newGrads;
+  v.numSkipped = s.newSkipped;
+  s.newBatch = s.newGrads = s.newSkipped = 0;
   uint64_t epoch = epoch_;
   std::weak_ptr<Accumulator> weak = weak_from_this();
-  group_->allReduce(fn("count"), std::move(v), nullptr,
-                    [weak, epoch](ReduceValue* rv, const std::string* err) {
-                      auto self = weak.lock();
-                      if (!self) return;
-                      std::lock_guard<std::mutex> lk(self->mu_);
-                      if (epoch != self->epoch_ || self->gradPhase_ != GradPhase::counting) return;
-                      if (err) {
-                        MRL_LOG_INFO("accumulator '%s': count round failed (%s); resync",
-                                     self->name_.c_str(), err->c_str());
-                        self->phase_ = Phase::inactive;
-                        self->syncSeen_ = 0;
-                        self->group_->requestResync();
-                        return;
-                      }
-                      self->totBatch_ += rv->batchSize;
-                      self->totGrads_ += rv->numGradients;
-                      self->totSkipped_ += rv->numSkipped;
-                      self->decided_ = false;
-                      if (self->totBatch_ >= self->virtualBatchSize_ && self->totGrads_ > 0) {
-                        self->startGradReduceLocked();
-                      } else {
-                        self->gradPhase_ = GradPhase::wantDecision;
-                      }
-                    });
+  group_->allReduce(
+      fn("count") + "/" + std::to_string(si), std::move(v), nullptr,
+      [weak, epoch, si](ReduceValue* rv, const std::string* err) {
+        auto self = weak.lock();
+        if (!self) return;
+        std::lock_guard<std::mutex> lk(self->mu_);
+        if (epoch != self->epoch_) return;
+        GradSlot& s = self->slots_[si];
+        if (s.phase != GradPhase::counting) return;
+        if (err) {
+          self->failAndResyncLocked(("count round failed: " + *err).c_str());
+          return;
+        }
+        s.totBatch += rv->batchSize;
+        s.totGrads += rv->numGradients;
+        s.totSkipped += rv->numSkipped;
+        self->decided_ = false;
+        if (s.totBatch >= self->virtualBatchSize_ && s.totGrads > 0) {
+          self->enterReducingLocked(si);
+        } else {
+          s.phase = GradPhase::wantDecision;
+        }
+      });
 }
 
-void Accumulator::startGradReduceLocked() {
-  gradPhase_ = GradPhase::reducing;
-  gradPhaseStarted_ = now();
+// The slot's virtual batch is full: start the gradient reduction and move
+// the cursor to the next free slot (pipelining; reductions against up to
+// slots_.size()-1 stale models, = moolib's set_parallel_gradients).
+void Accumulator::enterReducingLocked(size_t si) {
+  GradSlot& s = slots_[si];
+  s.phase = GradPhase::reducing;
+  s.started = now();
+  slotCursor_ = (si + 1) % slots_.size();
   if (hook_) {
-    if (hookPoll_) {
-      // Previous (abandoned) collective still draining: start ours once it
-      // completes (update() watches hookStartPending_).
-      hookStartPending_ = true;
-      return;
-    }
-    // MI355X fast path: in-place sum over the fixed torch.distributed world
-    // (RCCL over xGMI). Peers that skipped contribute zeros.
-    try {
-      hookPoll_ = hook_(flat_);
-    } catch (const std::exception& e) {
-      MRL_LOG_ERROR("local reduce hook failed: %s", e.what());
-      phase_ = Phase::inactive;
-      syncSeen_ = 0;
-      group_->requestResync();
-    }
+    // Launch order of collectives must be identical on every peer: slot
+    // transitions are (count rounds are sequential cluster-wide), so a
+    // ticket taken at transition time is globally consistent.
+    s.launchSeq = hookLaunchCounter_++;
+    s.hookPending = true;
+    tryLaunchHooksLocked();
     return;
   }
-  // RPC tree path: ship the flat bucket (staged to CPU) through the group
+  // RPC tree path: ship the bucket (staged to CPU) through the group
   // allreduce. Cross-node / elastic fallback.
   ReduceValue v;
   v.kind = ReduceValue::gradBundle;
-  v.tensors.push_back(flat_.device().is_cpu() ? flat_.clone() : flat_.to(at::kCPU));
+  {
+    NoGrad ng;
+    v.tensors.push_back(s.flat.device().is_cpu() ? s.flat.clone() : s.flat.to(at::kCPU));
+  }
   uint64_t epoch = epoch_;
   std::weak_ptr<Accumulator> weak = weak_from_this();
-  group_->allReduce(fn("grads"), std::move(v), nullptr,
-                    [weak, epoch](ReduceValue* rv, const std::string* err) {
-                      auto self = weak.lock();
-                      if (!self) return;
-                      std::lock_guard<std::mutex> lk(self->mu_);
-                      if (epoch != self->epoch_ || self->gradPhase_ != GradPhase::reducing) return;
-                      if (err) {
-                        MRL_LOG_INFO("accumulator '%s': gradient reduce failed (%s); resync",
-                                     self->name_.c_str(), err->c_str());
-                        self->phase_ = Phase::inactive;
-                        self->syncSeen_ = 0;
-                        self->group_->requestResync();
-                        return;
-                      }
-                      NoGrad ng;
-                      at::Tensor result = rv->tensors.at(0);
-                      if (!self->flat_.device().is_cpu()) {
-                        self->flat_.copy_(result, /*non_blocking=*/true);
-                        self->applyGradResultLocked(self->flat_);
-                      } else {
-                        // (Group::completeOp already cloned the tensors, so
-                        // in-place mutation here is safe.)
-                        self->applyGradResultLocked(result);
-                      }
-                    });
+  group_->allReduce(
+      fn("grads") + "/" + std::to_string(si), std::move(v), nullptr,
+      [weak, epoch, si](ReduceValue* rv, const std::string* err) {
+        auto self = weak.lock();
+        if (!self) return;
+        std::lock_guard<std::mutex> lk(self->mu_);
+        if (epoch != self->epoch_) return;
+        GradSlot& s = self->slots_[si];
+        if (s.phase != GradPhase::reducing) return;
+        if (err) {
+          self->failAndResyncLocked(("gradient reduce failed: " + *err).c_str());
+          return;
+        }
+        NoGrad ng;
+        at::Tensor result = rv->tensors.at(0);
+        if (!s.flat.device().is_cpu()) result = result.to(s.flat.device(), /*non_blocking=*/true);
+        self->completeSlotLocked(si, std::move(result));
+      });
 }
 
-void Accumulator::applyGradResultLocked(at::Tensor flatResult) {
+void Accumulator::tryLaunchHooksLocked() {
+  if (!hook_ || !drains_.empty()) return;
+  // Launch in ticket order; one per call is fine (update runs every iter).
+  for (size_t si = 0; si < slots_.size(); ++si) {
+    GradSlot& s = slots_[si];
+    if (s.phase == GradPhase::reducing && s.hookPending && s.launchSeq == hookLaunchedUpTo_) {
+      s.hookPending = false;
+      ++hookLaunchedUpTo_;
+      try {
+        s.hookPoll = hook_(s.flat);
+      } catch (const std::exception& e) {
+        MRL_LOG_ERROR("local reduce hook failed: %s", e.what());
+        failAndResyncLocked("hook launch failed");
+        return;
+      }
+    }
+  }
+}
+
+void Accumulator::completeSlotLocked(size_t si, at::Tensor result) {
+  GradSlot& s = slots_[si];
+  results_.push_back(PendingResult{std::move(result), s.totBatch, s.totGrads, s.totSkipped});
+  s.totBatch = s.totGrads = s.totSkipped = 0;
+  s.phase = GradPhase::wantDecision;
+  {
+    NoGrad ng;
+    s.flat.zero_();
+  }
+  // NOTE: the result is NOT applied here. Completion runs on a scheduler
+  // thread at an arbitrary point in the user's iteration; writing .grad now
+  // could clobber a backward() in progress (possible when parallel slots
+  // keep wants_gradients() true while reductions are in flight). Results
+  // apply only from update() / zero_gradients(), which the cooperative
+  // loop calls outside its backward/reduce block.
+}
+
+void Accumulator::applyPendingLocked() {
+  if (hasGradients_ || results_.empty()) return;
+  PendingResult r = std::move(results_.front());
+  results_.pop_front();
   NoGrad ng;
-  int64_t n = std::max<int64_t>(totGrads_, 1);
-  flatResult.div_(static_cast<double>(n));
+  int64_t n = std::max<int64_t>(r.grads, 1);
+  r.flat.div_(static_cast<double>(n));
   for (size_t i = 0; i < params_.size(); ++i) {
     auto& p = params_[i];
-    at::Tensor slice = flatResult.narrow(0, offsets_[i], numels_[i]).view(p.sizes());
+    at::Tensor slice = r.flat.narrow(0, offsets_[i], numels_[i]).view(p.sizes());
     if (!p.grad().defined()) {
       p.mutable_grad() = slice.clone();
     } else {
       p.grad().copy_(slice, /*non_blocking=*/true);
     }
   }
-  statBatch_ = totBatch_;
-  statGrads_ = totGrads_;
-  statSkipped_ = totSkipped_;
-  totBatch_ = totGrads_ = totSkipped_ = 0;
+  statBatch_ = r.batch;
+  statGrads_ = r.grads;
+  statSkipped_ = r.skipped;
   modelVersion_ += 1;
   hasGradients_ = true;
-  gradPhase_ = GradPhase::resultReady;
 }
 
 void Accumulator::maybeSendModelUpdatesLocked() {
@@ -481,8 +515,11 @@ std::pair<std::string, std::vector<at::Tensor>> Accumulator::state() {
 
 bool Accumulator::wantsGradients() {
   std::lock_guard<std::mutex> lk(mu_);
-  return phase_ == Phase::running && gradPhase_ == GradPhase::wantDecision && !decided_ &&
-         !hasGradients_;
+  if (phase_ != Phase::running || decided_) return false;
+  if (slots_[slotCursor_].phase != GradPhase::wantDecision) return false;
+  // Bound unapplied work: at most slots_.size() results outstanding.
+  size_t pending = results_.size() + (hasGradients_ ? 1 : 0);
+  return pending < slots_.size();
 }
 
 bool Accumulator::hasGradients() {
@@ -492,27 +529,30 @@ bool Accumulator::hasGradients() {
 
 void Accumulator::skipGradients() {
   std::lock_guard<std::mutex> lk(mu_);
-  if (phase_ != Phase::running || gradPhase_ != GradPhase::wantDecision || decided_) return;
+  if (phase_ != Phase::running || decided_) return;
+  GradSlot& s = slots_[slotCursor_];
+  if (s.phase != GradPhase::wantDecision) return;
   decided_ = true;
-  newSkipped_ += 1;
+  s.newSkipped += 1;
 }
 
 void Accumulator::reduceGradients(int64_t batchSize) {
   std::lock_guard<std::mutex> lk(mu_);
-  if (phase_ != Phase::running || gradPhase_ != GradPhase::wantDecision || decided_) {
+  GradSlot& s = slots_[slotCursor_];
+  if (phase_ != Phase::running || s.phase != GradPhase::wantDecision || decided_) {
     throw RpcError("reduce_gradients called when wants_gradients() is false");
   }
   NoGrad ng;
   for (size_t i = 0; i < params_.size(); ++i) {
     auto& p = params_[i];
     if (p.grad().defined()) {
-      flat_.narrow(0, offsets_[i], numels_[i]).add_(p.grad().flatten());
+      s.flat.narrow(0, offsets_[i], numels_[i]).add_(p.grad().flatten());
       p.grad().zero_();
     }
   }
   decided_ = true;
-  newBatch_ += batchSize;
-  newGrads_ += 1;
+  s.newBatch += batchSize;
+  s.newGrads += 1;
 }
 
 void Accumulator::zeroGradients() {
@@ -521,9 +561,8 @@ void Accumulator::zeroGradients() {
   for (auto& p : params_) {
     if (p.grad().defined()) p.grad().zero_();
   }
-  flat_.zero_();
   hasGradients_ = false;
-  if (gradPhase_ == GradPhase::resultReady) gradPhase_ = GradPhase::wantDecision;
+  applyPendingLocked();
 }
 
 int64_t Accumulator::modelVersion() {
@@ -558,7 +597,14 @@ void Accumulator::setVirtualBatchSize(int64_t n) {
 
 void Accumulator::setParallelGradients(int64_t n) {
   std::lock_guard<std::mutex> lk(mu_);
-  parallelGradients_ = std::max<int64_t>(n, 1);
+  n = std::max<int64_t>(n, 1);
+  if (static_cast<size_t>(n) == slots_.size()) return;
+  // Resize between rounds only; all members must configure the same value
+  // before training (mirrors the reference's usage).
+  slots_.resize(n);
+  for (auto& s : slots_) {
+    if (!s.flat.defined()) s.flat = makeFlatLocked();
+  }
 }
 
 void Accumulator::setLocalReduceHook(LocalReduceHook h) {
@@ -570,9 +616,13 @@ std::string Accumulator::debugState() {
   std::lock_guard<std::mutex> lk(mu_);
   std::ostringstream os;
   os << "Accumulator '" << name_ << "' phase=" << static_cast<int>(phase_)
-     << " gradPhase=" << static_cast<int>(gradPhase_) << " leader='" << leader_ << "'"
-     << " version=" << modelVersion_ << " decided=" << decided_ << " totBatch=" << totBatch_
-     << " hasGrads=" << hasGradients_;
+     << " leader='" << leader_ << "' version=" << modelVersion_ << " decided=" << decided_
+     << " cursor=" << slotCursor_ << " hasGrads=" << hasGradients_
+     << " results=" << results_.size() << " slots=[";
+  for (auto& s : slots_) {
+    os << static_cast<int>(s.phase) << "(tb=" << s.totBatch << ")";
+  }
+  os << "]";
   return os.str();
 }
 

@@ -18,6 +18,8 @@
 // allreduce (cross-node / elastic fallback).
 #pragma once
 
+#include <deque>
+
 #include "services.h"
 
 namespace mrl {
@@ -73,12 +75,31 @@ class Accumulator : public std::enable_shared_from_this<Accumulator> {
   enum class Phase { inactive, electing, fetching, running };
   enum class GradPhase { wantDecision, counting, reducing, resultReady };
 
+  struct GradSlot {
+    at::Tensor flat;  // on params' device; local accumulated gradient sum
+    int64_t newBatch = 0, newGrads = 0, newSkipped = 0;   // since last count round
+    int64_t totBatch = 0, totGrads = 0, totSkipped = 0;   // accumulated global totals
+    GradPhase phase = GradPhase::wantDecision;
+    std::function<bool()> hookPoll;
+    uint64_t launchSeq = 0;     // hook-launch ordering ticket
+    bool hookPending = false;   // waiting for its launch turn / drains
+    TimePoint started{};
+  };
+  struct PendingResult {
+    at::Tensor flat;  // globally summed gradients (on device)
+    int64_t batch, grads, skipped;
+  };
+
   // all called with mu_ held unless noted
   void resetLocked(const char* why);
   void startElectionLocked();
   void startCountRoundLocked();
-  void startGradReduceLocked();
-  void applyGradResultLocked(at::Tensor flatResult);
+  void enterReducingLocked(size_t si);
+  void tryLaunchHooksLocked();
+  void completeSlotLocked(size_t si, at::Tensor result);
+  void applyPendingLocked();
+  void failAndResyncLocked(const char* why);
+  at::Tensor makeFlatLocked();
   void maybeSendModelUpdatesLocked();
   std::string fn(const char* suffix) const { return "__mrl_acc_" + std::string(suffix) + ":" + name_; }
 
@@ -112,21 +133,27 @@ class Accumulator : public std::enable_shared_from_this<Accumulator> {
   std::vector<std::string> stateRequesters_;
   TimePoint lastBuffersBroadcast_{};
 
-  // gradient machine
-  at::Tensor flat_;          // on params' device; local accumulated gradient sum
+  // gradient machine: parallelGradients_ staging slots, used round-robin.
+  // Count rounds are strictly sequential cluster-wide (decided via shared
+  // allreduce results), so every member walks the same slot/round sequence;
+  // a slot entering `reducing` frees the cursor to the next slot, bounding
+  // in-flight reductions to the slot count (moolib's set_parallel_gradients
+  // pipelining, reference accumulator.cc:251-256).
+  std::vector<GradSlot> slots_;
+  size_t slotCursor_ = 0;
   std::vector<int64_t> offsets_, numels_;
   int64_t virtualBatchSize_ = 1;
-  int64_t parallelGradients_ = 1;
   bool decided_ = false;
-  int64_t newBatch_ = 0, newGrads_ = 0, newSkipped_ = 0;   // since last count round
-  int64_t totBatch_ = 0, totGrads_ = 0, totSkipped_ = 0;   // accumulated global totals
   bool hasGradients_ = false;
   int64_t statBatch_ = 0, statGrads_ = 0, statSkipped_ = 0;  // of the applied result
+  std::deque<PendingResult> results_;
   LocalReduceHook hook_;
-  std::function<bool()> hookPoll_;
-  bool hookAbandoned_ = false;     // in-flight collective whose round we left
-  bool hookStartPending_ = false;  // next collective waits for the drain
-  TimePoint gradPhaseStarted_{};
+  uint64_t hookLaunchCounter_ = 0;   // tickets handed to reducing slots
+  uint64_t hookLaunchedUpTo_ = 0;    // tickets actually launched
+  // Collectives whose round was reset: polled to completion so the peers'
+  // call sequences stay aligned; their (mixed-round) output buffers are
+  // owned by the closure and discarded.
+  std::vector<std::function<bool()>> drains_;
 };
 
 }  // namespace mrl

@@ -279,3 +279,55 @@ class TestDistributedDataPlane:
             assert p.exitcode == 0, "worker failed with %s" % p.exitcode
         for r in range(world):
             assert (tmp_path / ("ok%d" % r)).exists()
+
+
+class TestParallelGradients:
+    def test_pipelined_rounds(self):
+        """set_parallel_gradients(2): round k+1 can be contributed while
+        round k's reduction is in flight; results apply in round order."""
+        c = AccCluster(2)
+        for p in c.peers:
+            p.acc.set_parallel_gradients(2)
+            p.acc.set_virtual_batch_size(1)
+        c.wait_connected()
+
+        rounds = 5
+        contributed = [0, 0]
+        applied = [[] for _ in range(2)]
+        overlapped = [False]
+
+        def step():
+            c.broker.update()
+            for i, p in enumerate(c.peers):
+                p.pump_once()
+                # canonical cooperative order: consume results BEFORE
+                # producing new gradients (matches the reference loop)
+                if p.acc.has_gradients():
+                    stats = p.acc.get_gradient_stats()
+                    assert stats["num_gradients"] == 2, stats
+                    applied[i].append(p.params[0].grad[0, 0].item())
+                    p.acc.zero_gradients()
+                elif p.acc.wants_gradients() and contributed[i] < rounds:
+                    val = float(10 * contributed[i] + i)
+                    for t in p.params:
+                        t.grad = torch.full_like(t, val)
+                    p.acc.reduce_gradients(1)
+                    contributed[i] += 1
+                    # pipelining: the other slot still has a round in flight
+                    if "1(" in p.acc.debug_state() or "2(" in p.acc.debug_state():
+                        overlapped[0] = True
+
+        t0 = time.time()
+        while (
+            min(len(applied[0]), len(applied[1])) < rounds and time.time() - t0 < 40
+        ):
+            step()
+            time.sleep(0.002)
+
+        want = sorted(10 * k + 0.5 for k in range(rounds))
+        for i in range(2):
+            assert len(applied[i]) >= rounds, (applied, [p.acc.debug_state() for p in c.peers])
+            # results may complete slightly out of round order across slots;
+            # the multiset of applied averages must match exactly
+            assert sorted(applied[i][:rounds]) == want, applied[i]
+        assert overlapped[0], "no pipelining observed"
