@@ -198,11 +198,11 @@ class TestAccumulator:
         assert p.acc.model_version() == v0 + 1
 
 
-def _dist_worker(rank, world, port, results_dir):
+def _dist_worker(rank, world, port, results_dir, parallel=1, rounds=1):
     import torch.distributed as dist
 
     import moolib_amd
-    from moolib_amd import parallel
+    from moolib_amd import parallel as parallel_mod
 
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
@@ -222,13 +222,16 @@ def _dist_worker(rank, world, port, results_dir):
     group.set_timeout(10)
     params, buffers = make_model(seed=37 + rank)
     acc = moolib_amd.Accumulator("acc", params, buffers, group=group)
+    if parallel > 1:
+        acc.set_parallel_gradients(parallel)
+        acc.set_virtual_batch_size(1)
     acc.connect("127.0.0.1:%d" % (port + 1))
-    parallel.install_collective_backend(acc)
+    parallel_mod.install_collective_backend(acc)
 
     t0 = time.time()
-    applied = False
-    contributed = False
-    while time.time() - t0 < 60 and not applied:
+    applied = 0
+    contributed = 0
+    while time.time() - t0 < 60 and applied < rounds:
         if broker:
             broker.update()
         acc.update()
@@ -237,11 +240,6 @@ def _dist_worker(rank, world, port, results_dir):
         if acc.has_new_state():
             acc.state()
         if acc.connected() and len(group.members()) == world:
-            if acc.wants_gradients() and not contributed:
-                for t in params:
-                    t.grad = torch.full_like(t, float(rank + 1))
-                acc.reduce_gradients(4)
-                contributed = True
             if acc.has_gradients():
                 stats = acc.get_gradient_stats()
                 expect = sum(r + 1 for r in range(world)) / world
@@ -250,10 +248,17 @@ def _dist_worker(rank, world, port, results_dir):
                         t.grad, torch.full_like(t, expect)
                     ), "rank %d wrong grads" % rank
                 assert stats["num_gradients"] == world
-                applied = True
+                acc.zero_gradients()
+                applied += 1
+            elif acc.wants_gradients() and contributed < rounds:
+                for t in params:
+                    t.grad = torch.full_like(t, float(rank + 1))
+                acc.reduce_gradients(4)
+                contributed += 1
         time.sleep(0.005)
 
-    assert applied, "rank %d never applied gradients: %s" % (rank, acc.debug_state())
+    assert applied >= rounds, "rank %d applied %d/%d: %s" % (
+        rank, applied, rounds, acc.debug_state())
     with open(os.path.join(results_dir, "ok%d" % rank), "w") as f:
         f.write("ok")
     dist.barrier()
@@ -261,15 +266,19 @@ def _dist_worker(rank, world, port, results_dir):
 
 
 class TestDistributedDataPlane:
-    def test_gloo_collective_hook(self, tmp_path):
-        """2 ranks, gloo backend: same code path bench.py uses with RCCL."""
+    @pytest.mark.parametrize("parallel,rounds", [(1, 1), (2, 4)])
+    def test_gloo_collective_hook(self, tmp_path, parallel, rounds):
+        """2 ranks, gloo backend: same code path bench.py uses with RCCL.
+        parallel=2 covers pipelined collective launches + ticket ordering."""
         import torch.multiprocessing as mp
 
-        port = 29000 + (os.getpid() % 500)
+        port = 29000 + (os.getpid() % 450) + parallel * 37
         world = 2
         ctx = mp.get_context("spawn")
         procs = [
-            ctx.Process(target=_dist_worker, args=(r, world, port, str(tmp_path)))
+            ctx.Process(
+                target=_dist_worker, args=(r, world, port, str(tmp_path), parallel, rounds)
+            )
             for r in range(world)
         ]
         for p in procs:
