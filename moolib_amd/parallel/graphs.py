@@ -54,7 +54,13 @@ class GraphedCall:
                     # dedicated (non-default) generators must be registered
                     # before capture so replays advance their offsets
                     g.register_generator_state(gen.graphsafe_get_state())
-                with torch.cuda.graph(g):
+                # thread_local: background threads (the accumulator's
+                # scheduler staging gradient buckets, RPC callbacks) keep
+                # doing legitimate GPU work while we capture; the default
+                # global mode lets their ops interleave into the capture,
+                # which intermittently produced corrupted graphs that fault
+                # at replay (HSA_STATUS_ERROR_EXCEPTION ~1 in 8 runs).
+                with torch.cuda.graph(g, capture_error_mode="thread_local"):
                     self.static_out = self.fn(self.static_in)
                 self.graph = g
                 logging.info("captured hipGraph '%s'", self.name)
