@@ -195,6 +195,10 @@ class ImpalaPeer:
         # env kill-switches for debugging/bisection
         if os.environ.get("MOOLIB_AMD_NO_GRAPHS"):
             cfg.graph_actor = cfg.graph_learner = False
+        if os.environ.get("MOOLIB_AMD_NO_ACTOR_GRAPH"):
+            cfg.graph_actor = False
+        if os.environ.get("MOOLIB_AMD_NO_LEARNER_GRAPH"):
+            cfg.graph_learner = False
         if os.environ.get("MOOLIB_AMD_NO_SIDE_STREAM"):
             cfg.actor_side_stream = False
         if os.environ.get("MOOLIB_AMD_FORCE_SIDE_STREAM"):
