@@ -41,6 +41,10 @@ def action_log_probs(policy_logits, actions):
 
 
 def _kernels():
+    import os
+
+    if os.environ.get("MOOLIB_AMD_NO_VTRACE_KERNEL"):
+        return None
     try:
         from moolib_amd import _kernels as k
 
