@@ -204,6 +204,8 @@ class ImpalaPeer:
             cfg.graph_actor = False
         if os.environ.get("MOOLIB_AMD_NO_LEARNER_GRAPH"):
             cfg.graph_learner = False
+        if os.environ.get("MOOLIB_AMD_FORCE_LEARNER_GRAPH"):
+            cfg.graph_learner = True
         if os.environ.get("MOOLIB_AMD_NO_SIDE_STREAM"):
             cfg.actor_side_stream = False
         if os.environ.get("MOOLIB_AMD_FORCE_SIDE_STREAM"):
