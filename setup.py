@@ -8,7 +8,6 @@ Two in-tree extensions:
 Usage: python setup.py build_ext --inplace
 """
 import os
-import sys
 from pathlib import Path
 
 from setuptools import setup

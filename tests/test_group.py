@@ -6,7 +6,6 @@ consistency) against our implementation.
 """
 import time
 
-import pytest
 import torch
 
 import moolib_amd

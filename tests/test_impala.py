@@ -6,7 +6,6 @@ V-trace gradients, Accumulator reduction, optimizer stepping.
 import time
 
 import pytest
-import torch
 
 import moolib_amd
 from moolib_amd.envs import SyntheticAtariEnv
