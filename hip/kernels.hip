@@ -279,6 +279,62 @@ __global__ void frames_u8_to_bf16_nhwc_kernel(const uint8_t* __restrict__ in,
   out[tid] = (hip_bfloat16)((float)v * scale);
 }
 
+// ------------------------------------------ fused first conv (actor path)
+//
+// conv1 of the IMPALA ResNet: uint8 NCHW frames -> 3x3 conv (C_in=4,
+// C_out=16, pad 1) + bias -> bf16 NHWC, in ONE kernel. At C_in=4 the
+// implicit-GEMM K is 36 — far too small for MFMA to pay — and the op is
+// VALU/L1-bound, so a direct conv that also folds the uint8 scale (and
+// skips the separate preprocessing pass + its 19 MB intermediate) wins.
+// One thread per output pixel computes all 16 output channels; the 1.2 KB
+// weight block is staged in LDS.
+
+__global__ void conv1_u8_nhwc_kernel(const uint8_t* __restrict__ in,   // [N,4,H,W]
+                                     const hip_bfloat16* __restrict__ w,  // [3,3,4,16]
+                                     const hip_bfloat16* __restrict__ bias,  // [16]
+                                     hip_bfloat16* __restrict__ out,  // [N,H,W,16] (NHWC)
+                                     float scale, int N, int H, int W) {
+  constexpr int CI = 4, CO = 16;
+  __shared__ float wsm[3 * 3 * CI * CO];
+  __shared__ float bsm[CO];
+  for (int i = threadIdx.x; i < 3 * 3 * CI * CO; i += blockDim.x) wsm[i] = (float)w[i];
+  for (int i = threadIdx.x; i < CO; i += blockDim.x) bsm[i] = (float)bias[i];
+  __syncthreads();
+
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)N * H * W;
+  if (tid >= total) return;
+  int ow = tid % W;
+  int64_t t = tid / W;
+  int oh = t % H;
+  int n = t / H;
+
+  float acc[CO];
+#pragma unroll
+  for (int co = 0; co < CO; ++co) acc[co] = bsm[co];
+
+#pragma unroll
+  for (int kh = 0; kh < 3; ++kh) {
+    int ih = oh + kh - 1;
+    if (ih < 0 || ih >= H) continue;
+#pragma unroll
+    for (int kw = 0; kw < 3; ++kw) {
+      int iw = ow + kw - 1;
+      if (iw < 0 || iw >= W) continue;
+      const float* wp = &wsm[(kh * 3 + kw) * CI * CO];
+#pragma unroll
+      for (int ci = 0; ci < CI; ++ci) {
+        float xv = (float)in[(((int64_t)n * CI + ci) * H + ih) * W + iw] * scale;
+#pragma unroll
+        for (int co = 0; co < CO; ++co) acc[co] = fmaf(xv, wp[ci * CO + co], acc[co]);
+      }
+    }
+  }
+  hip_bfloat16* op = out + tid * CO;
+#pragma unroll
+  for (int co = 0; co < CO; ++co) op[co] = (hip_bfloat16)acc[co];
+}
+
 }  // namespace
 
 #include "lstm.hip.inc"
@@ -410,6 +466,27 @@ at::Tensor frames_u8_to_bf16_nhwc(at::Tensor x, double scale) {
   return out;
 }
 
+at::Tensor conv1_u8_nhwc(at::Tensor x, at::Tensor w, at::Tensor bias, double scale) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == at::kByte && x.dim() == 4 && x.size(1) == 4,
+              "conv1: uint8 [N,4,H,W] expected");
+  TORCH_CHECK(w.dtype() == at::kBFloat16 && w.numel() == 3 * 3 * 4 * 16,
+              "conv1: packed bf16 [3,3,4,16] weights expected");
+  auto xc = x.contiguous();
+  int N = xc.size(0), H = xc.size(2), W = xc.size(3);
+  auto out = at::empty({N, 16, H, W},
+                       xc.options().dtype(at::kBFloat16).memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * H * W;
+  int threads = 256;
+  int64_t blocks = (total + threads - 1) / threads;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  hipLaunchKernelGGL(conv1_u8_nhwc_kernel, dim3(blocks), dim3(threads), 0, stream,
+                     xc.data_ptr<uint8_t>(),
+                     reinterpret_cast<const hip_bfloat16*>(w.contiguous().data_ptr()),
+                     reinterpret_cast<const hip_bfloat16*>(bias.contiguous().data_ptr()),
+                     reinterpret_cast<hip_bfloat16*>(out.data_ptr()), (float)scale, N, H, W);
+  return out;
+}
+
 std::vector<at::Tensor> maxpool3x3s2_fwd(at::Tensor x) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4, "maxpool: 4D CUDA tensor expected");
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "maxpool: channels_last expected");
@@ -457,6 +534,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd, "NHWC 3x3/2 maxpool forward (gfx950)");
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd, "NHWC 3x3/2 maxpool backward (gather, no atomics)");
   m.def("frames_u8_to_bf16_nhwc", &frames_u8_to_bf16_nhwc, "fused uint8->bf16 NHWC scale");
+  m.def("conv1_u8_nhwc", &conv1_u8_nhwc, "fused uint8 frames -> conv(4->16,3x3)+bias, NHWC bf16");
   m.def("lstm_fused_fwd", &lstm_fused_fwd, "fused masked LSTM sequence scan fwd (MFMA, gfx950)");
   m.def("lstm_fused_bwd", &lstm_fused_bwd, "fused masked LSTM sequence scan bwd (MFMA, gfx950)");
   m.def("vtrace_from_log_rhos", &vtrace_from_log_rhos, "fused V-trace scan (gfx950)");

@@ -86,22 +86,36 @@ class AtariNet(nn.Module):
         reward = inputs["reward"]
         T, B = x.shape[:2]
         x = x.flatten(0, 1)
-        fused = None
         import os as _os
 
-        if x.is_cuda and x.dtype == torch.uint8 and not _os.environ.get(
-            "MOOLIB_AMD_NO_FRAMES_KERNEL"
-        ) and (
-            torch.is_autocast_enabled() or self.fc.weight.dtype == torch.bfloat16
-        ):
+        bf16_compute = torch.is_autocast_enabled() or self.fc.weight.dtype == torch.bfloat16
+        kernels = None
+        if x.is_cuda and x.dtype == torch.uint8 and bf16_compute:
             try:
-                from moolib_amd import _kernels
-
-                fused = _kernels.frames_u8_to_bf16_nhwc(x, 1.0 / 255.0)
+                from moolib_amd import _kernels as kernels
             except ImportError:
-                fused = None
-        if fused is not None:
-            x = fused
+                kernels = None
+
+        first = 0
+        if (
+            kernels is not None
+            and not torch.is_grad_enabled()  # actor path; learner conv needs autograd
+            and self.sections[0].conv.in_channels == 4
+            and not _os.environ.get("MOOLIB_AMD_NO_CONV1_KERNEL")
+        ):
+            # Fused uint8 frames -> conv1 + bias in one kernel (no separate
+            # preprocessing pass, no 19 MB fp intermediate).
+            c1 = self.sections[0].conv
+            w = c1.weight.detach().to(torch.bfloat16).permute(2, 3, 1, 0).contiguous()
+            b = c1.bias.detach().to(torch.bfloat16)
+            x = kernels.conv1_u8_nhwc(x, w, b, 1.0 / 255.0)
+            from moolib_amd.ops.pool import maxpool3x3s2
+
+            sec0 = self.sections[0]
+            x = sec0.res1(sec0.res0(maxpool3x3s2(x)))
+            first = 1
+        elif kernels is not None and not _os.environ.get("MOOLIB_AMD_NO_FRAMES_KERNEL"):
+            x = kernels.frames_u8_to_bf16_nhwc(x, 1.0 / 255.0)
         else:
             x = x.to(self.fc.weight.dtype).mul_(1.0 / 255.0)
             if x.is_cuda:
@@ -112,7 +126,7 @@ class AtariNet(nn.Module):
                 # features.)
                 x = x.contiguous(memory_format=torch.channels_last)
 
-        for s in self.sections:
+        for s in self.sections[first:]:
             x = s(x)
         x = F.relu(x)
         x = x.reshape(T * B, -1)

@@ -307,3 +307,46 @@ class TestFusedLSTM:
             out_fused["policy_logits"], out_loop["policy_logits"], atol=0.15
         ), (out_fused["policy_logits"] - out_loop["policy_logits"]).abs().max()
         assert torch.allclose(out_fused["baseline"], out_loop["baseline"], atol=0.15)
+
+
+@gpu
+@requires_gpu
+class TestConv1Kernel:
+    def test_matches_conv2d(self):
+        from moolib_amd import _kernels
+
+        torch.manual_seed(9)
+        conv = torch.nn.Conv2d(4, 16, 3, stride=1, padding=1).cuda()
+        x = torch.randint(0, 256, (6, 4, 84, 84), dtype=torch.uint8, device="cuda")
+        w = conv.weight.detach().to(torch.bfloat16).permute(2, 3, 1, 0).contiguous()
+        b = conv.bias.detach().to(torch.bfloat16)
+        got = _kernels.conv1_u8_nhwc(x, w, b, 1.0 / 255.0)
+        with torch.no_grad():
+            want = conv(x.float() / 255.0)
+        assert got.is_contiguous(memory_format=torch.channels_last)
+        d = (got.float() - want).abs().max().item()
+        assert d < 0.03, d  # bf16 weights vs fp32 reference
+
+    def test_actor_model_path_consistent(self):
+        """AtariNet no-grad forward: fused-conv1 path vs frames-kernel path."""
+        import os
+
+        from moolib_amd.models.atari import AtariNet
+
+        torch.manual_seed(10)
+        model = AtariNet(num_actions=6).cuda().to(torch.bfloat16)
+        inputs = {
+            "state": torch.randint(0, 256, (1, 16, 4, 84, 84), dtype=torch.uint8, device="cuda"),
+            "reward": torch.randn(1, 16, device="cuda"),
+            "done": torch.zeros(1, 16, dtype=torch.bool, device="cuda"),
+            "prev_action": torch.zeros(1, 16, dtype=torch.int64, device="cuda"),
+        }
+        with torch.no_grad():
+            out_fused, _ = model(inputs, tuple())
+            os.environ["MOOLIB_AMD_NO_CONV1_KERNEL"] = "1"
+            try:
+                out_plain, _ = model(inputs, tuple())
+            finally:
+                del os.environ["MOOLIB_AMD_NO_CONV1_KERNEL"]
+        d = (out_fused["policy_logits"] - out_plain["policy_logits"]).abs().max().item()
+        assert d < 0.25, d  # bf16 accumulation-order differences through the net
