@@ -212,6 +212,8 @@ class ImpalaPeer:
             cfg.actor_side_stream = True
         if os.environ.get("MOOLIB_AMD_BF16_WEIGHTS"):
             cfg.bf16_weights = True
+        if os.environ.get("MOOLIB_AMD_NO_BF16_WEIGHTS"):
+            cfg.bf16_weights = False
         self.autocast = cfg.autocast_bf16 and self.is_cuda
         if self.is_cuda and cfg.channels_last:
             self.model.to(memory_format=torch.channels_last)
