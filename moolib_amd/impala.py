@@ -357,8 +357,20 @@ class ImpalaPeer:
             cfg.baseline_cost,
         )
         total.backward()
+        return ()
+
+    def compute_gradients(self, data):
+        self.fwd_model.train()
+        # Shadow-grad bookkeeping stays OUTSIDE the captured region: zeroing
+        # and cast-copying the shadow grads inside the learner graph is what
+        # made replays fault on ROCm 7.0 (captured pure fwd+bwd is stable).
         if self.bf16_shadow:
-            # cast this batch's bf16 gradients onto the fp32 masters
+            with torch.no_grad():
+                grads = [pb.grad for pb in self._fwd_params if pb.grad is not None]
+                if grads:
+                    torch._foreach_zero_(grads)
+        self._learn_call(data)
+        if self.bf16_shadow:
             with torch.no_grad():
                 for pf, pb in zip(self._master_params, self._fwd_params):
                     if pb.grad is None:
@@ -367,11 +379,6 @@ class ImpalaPeer:
                         pf.grad = pb.grad.float()
                     else:
                         pf.grad.copy_(pb.grad, non_blocking=True)
-        return ()
-
-    def compute_gradients(self, data):
-        self.fwd_model.train()
-        self._learn_call(data)
         self.stats["env_train_steps"] += self.cfg.unroll_length * self.cfg.batch_size
 
     def _actor_fn(self, inputs):
