@@ -206,6 +206,18 @@ class PyQueue {
     std::mutex mu;
     std::deque<py::object> items;          // GIL
     std::deque<FutureStatePtr> waiters;
+    ~State() {
+      // May be destroyed from a non-Python thread (e.g. a define_queue
+      // handler closure cleared during Rpc::shutdown).
+      if (!items.empty()) {
+        if (pyAlive()) {
+          py::gil_scoped_acquire gil;
+          items.clear();
+        } else {
+          for (auto& o : items) (void)o.release();
+        }
+      }
+    }
   };
   PyQueue() : st_(std::make_shared<State>()) {}
 

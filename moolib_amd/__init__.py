@@ -11,6 +11,8 @@ import asyncio
 import atexit
 import threading
 
+__version__ = "0.1.0"
+
 import torch
 
 from . import _core
