@@ -422,6 +422,7 @@ void Group::advance(const OpKey& key, AllReduceOpPtr op) {
     {
       std::lock_guard<std::mutex> lk(mu_);
       if (op->completed || op->folding || !op->localContributed || op->childrenExpected < 0) return;
+      if (op->sentUp) return;  // late contribution: acc may be serializing on another thread
       if (!op->queued.empty()) {
         batch = std::move(op->queued);
         op->queued.clear();
