@@ -16,6 +16,22 @@ __version__ = "0.1.0"
 import torch
 
 from . import _core
+
+# The gfx950 kernel extension must load wherever a GPU is present: a silent
+# fallback to eager torch on an MI355X would defeat the point of the
+# framework (the ops modules only use eager paths off-GPU / in explicitly
+# disabled modes).
+try:
+    from . import _kernels  # noqa: F401
+except ImportError as _e:  # pragma: no cover
+    if torch.cuda.is_available():
+        raise ImportError(
+            "moolib_amd._kernels (gfx950 HIP kernels) failed to load on a GPU "
+            "machine: %s — rebuild with `python setup.py build_ext --inplace`"
+            % _e
+        ) from _e
+    _kernels = None
+
 from ._core import (
     Broker,
     Future,
