@@ -1,0 +1,64 @@
+"""RPC micro-benchmarks (measured, loosely asserted).
+
+Mirrors the reference's test/unit/test_tensors.py throughput smokes: sync
+noop latency and a large-tensor RPC (zero-copy tensor frames over loopback
+TCP).
+"""
+import time
+
+import torch
+
+import moolib_amd
+
+
+def make_pair():
+    host = moolib_amd.Rpc()
+    client = moolib_amd.Rpc()
+    host.set_name("host")
+    client.set_name("client")
+    client.set_timeout(60)
+    addr = host.listen("127.0.0.1:0")[0]
+    client.connect(addr)
+    return host, client
+
+
+class TestRpcThroughput:
+    def test_sync_noop_rate(self):
+        host, client = make_pair()
+        host.define("noop", lambda: None)
+        client.sync("host", "noop")  # connection warmup
+        n = 200
+        t0 = time.time()
+        for _ in range(n):
+            client.sync("host", "noop")
+        dt = time.time() - t0
+        rate = n / dt
+        print("sync noop: %.0f calls/s (%.2f ms each)" % (rate, 1000 * dt / n))
+        assert rate > 300, rate  # loose floor; typical is thousands
+
+    def test_async_noop_pipeline(self):
+        host, client = make_pair()
+        host.define("noop", lambda: None)
+        client.sync("host", "noop")
+        n = 2000
+        t0 = time.time()
+        futs = [client.async_("host", "noop") for _ in range(n)]
+        for f in futs:
+            f.result()
+        dt = time.time() - t0
+        print("async noop: %.0f calls/s" % (n / dt))
+        assert n / dt > 2000, n / dt
+
+    def test_large_tensor_bandwidth(self):
+        host, client = make_pair()
+        host.define("echo_sum", lambda t: t.sum())
+        x = torch.randn(4096, 4096)  # 64 MiB
+        client.sync("host", "echo_sum", x)  # warmup
+        n = 5
+        t0 = time.time()
+        for _ in range(n):
+            client.sync("host", "echo_sum", x)
+        dt = time.time() - t0
+        gbps = n * x.nbytes / dt / 1e9
+        print("64MiB tensor rpc: %.2f GB/s one-way payload" % gbps)
+        assert gbps > 0.2, gbps  # loose floor
