@@ -79,6 +79,15 @@ void setTcpOpts(int fd) {
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
 }
 
+// Large tensor frames want deep socket buffers (loopback included): the
+// defaults (~200 KB) throttle 64 MB gradient/model messages to well under
+// a GB/s.
+void setBufSizes(int fd) {
+  int sz = 8 << 20;
+  setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+  setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+}
+
 sockaddr_un makeUnixAddr(const std::string& name, socklen_t* len) {
   sockaddr_un sa{};
   sa.sun_family = AF_UNIX;
@@ -364,6 +373,7 @@ struct SocketEngine::Impl {
         return;
       }
       if (!isUnix) setTcpOpts(fd);
+      setBufSizes(fd);
       ConnId id = nextId.fetch_add(1);
       Conn& c = conns[id];
       c.fd = fd;
@@ -504,6 +514,7 @@ ConnId SocketEngine::connect(const std::string& addrStr) {
       return;
     }
     if (!isUnix) setTcpOpts(fd);
+    setBufSizes(fd);
     int rc = ::connect(fd, reinterpret_cast<const sockaddr*>(&ss), slen);
     if (rc != 0 && errno != EINPROGRESS) {
       std::string reason = std::string("connect failed: ") + strerror(errno);
