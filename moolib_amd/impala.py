@@ -59,6 +59,9 @@ class ImpalaConfig:
     lr_schedule: bool = True
     channels_last: bool = True   # NHWC convs (MIOpen direct, no transposes)
     graph_actor: bool = True     # hipGraph-capture the actor forward
+    graph_optimizer: bool = True  # hipGraph-capture clip_grad_norm + Adam step
+    #   (no autograd in the capture; only used when the LR is constant —
+    #   python-float hyperparameters are baked into the graph)
     graph_learner: bool = False  # hipGraph-capture the learner fwd+bwd.
     #   OFF by default: replays of the captured learner (autograd backward
     #   incl. MIOpen bwd-weight convs) intermittently fault on ROCm 7.0
@@ -177,6 +180,17 @@ class ImpalaPeer:
             )
         else:
             self.scheduler = None
+        self._graph_opt = (
+            cfg.graph_optimizer
+            and self.scheduler is None
+            and torch.device(cfg.device).type == "cuda"
+            and not os.environ.get("MOOLIB_AMD_NO_OPT_GRAPH")
+        )
+        if self._graph_opt:
+            # capturable: the Adam step counter must live on-device so graph
+            # replays advance bias correction
+            for g in self.optimizer.param_groups:
+                g["capturable"] = True
 
         self.rpc = moolib_amd.Rpc()
         self.rpc.set_name(cfg.local_name or ("peer-" + moolib_amd.create_uid()[:8]))
@@ -252,6 +266,12 @@ class ImpalaPeer:
             self._actor_rng = None
             self._learn_rng = None
         from moolib_amd.parallel.graphs import GraphedCall
+
+        self._opt_call = (
+            GraphedCall(self._opt_fn, warmup=3, name="optimizer_step")
+            if self._graph_opt
+            else self._opt_fn
+        )
 
         self._actor_call = (
             GraphedCall(
@@ -394,18 +414,24 @@ class ImpalaPeer:
             )
         return {"out": actor_outputs, "core": core_out if core_out else ()}
 
-    def step_optimizer(self):
-        cfg = self.cfg
-        norm = torch.nn.utils.clip_grad_norm_(self.model.parameters(), cfg.grad_norm_clipping)
+    def _opt_fn(self, _inputs):
+        norm = torch.nn.utils.clip_grad_norm_(
+            self.model.parameters(), self.cfg.grad_norm_clipping
+        )
         self.optimizer.step()
         self._sync_shadow()
+        return norm
+
+    def step_optimizer(self):
+        cfg = self.cfg
+        norm = self._opt_call({})
         if self.scheduler is not None:
             self.scheduler.step()
         self.model_version += 1
         # .item() is a device sync: sample the norm stat instead of paying
         # the sync every step.
         if not self.is_cuda or self.model_version % 16 == 0:
-            self.stats["unclipped_grad_norm"] += norm.item()
+            self.stats["unclipped_grad_norm"] += float(norm)
         self.stats["optimizer_steps"] += 1
         self.stats["model_version"] += 1
 
