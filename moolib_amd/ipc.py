@@ -20,8 +20,25 @@ driver).
 Lifetime: the producer's storage stays alive until every consumer drops
 its alias (torch's IPC refcounting). Do not send handles across nodes.
 """
+import os
+import weakref
+
 import torch
 from torch.multiprocessing import reductions
+
+# Same-process short-circuit: a process cannot hipIpcOpenMemHandle its own
+# handle, so handles deserialized by the producing process resolve through
+# this registry instead.
+_local = weakref.WeakValueDictionary()
+_counter = [0]
+
+
+def _materialize(pid, key, func, args):
+    if pid == os.getpid():
+        t = _local.get(key)
+        if t is not None:
+            return t
+    return func(*args)
 
 
 class SharedCudaTensor:
@@ -34,7 +51,10 @@ class SharedCudaTensor:
 
     def __reduce__(self):
         func, args = reductions.reduce_tensor(self._tensor)
-        return (func, args)
+        _counter[0] += 1
+        key = _counter[0]
+        _local[key] = self._tensor
+        return (_materialize, (os.getpid(), key, func, args))
 
     def tensor(self):
         return self._tensor
