@@ -21,23 +21,32 @@ Lifetime: the producer's storage stays alive until every consumer drops
 its alias (torch's IPC refcounting). Do not send handles across nodes.
 """
 import os
-import weakref
+import time
 
 import torch
 from torch.multiprocessing import reductions
 
 # Same-process short-circuit: a process cannot hipIpcOpenMemHandle its own
 # handle, so handles deserialized by the producing process resolve through
-# this registry instead.
-_local = weakref.WeakValueDictionary()
+# this registry instead. Strong refs (the producer must keep the storage
+# alive for consumers regardless); a TTL bounds never-consumed entries.
+_local = {}  # key -> (tensor, deadline)
 _counter = [0]
+_TTL = 120.0
+
+
+def _evict():
+    now = time.monotonic()
+    dead = [k for k, (_, dl) in _local.items() if dl < now]
+    for k in dead:
+        del _local[k]
 
 
 def _materialize(pid, key, func, args):
     if pid == os.getpid():
-        t = _local.get(key)
-        if t is not None:
-            return t
+        ent = _local.get(key)
+        if ent is not None:
+            return ent[0]
     return func(*args)
 
 
@@ -51,9 +60,10 @@ class SharedCudaTensor:
 
     def __reduce__(self):
         func, args = reductions.reduce_tensor(self._tensor)
+        _evict()
         _counter[0] += 1
         key = _counter[0]
-        _local[key] = self._tensor
+        _local[key] = (self._tensor, time.monotonic() + _TTL)
         return (_materialize, (os.getpid(), key, func, args))
 
     def tensor(self):
