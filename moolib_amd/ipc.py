@@ -44,7 +44,9 @@ def _evict():
 
 def _materialize(pid, key, func, args):
     if pid == os.getpid():
-        ent = _local.get(key)
+        # Pop: the consumer takes ownership, so serving a stream of fresh
+        # batches in-process doesn't pin them all for the TTL.
+        ent = _local.pop(key, None)
         if ent is not None:
             return ent[0]
     return func(*args)
