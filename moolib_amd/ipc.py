@@ -28,18 +28,35 @@ from torch.multiprocessing import reductions
 
 # Same-process short-circuit: a process cannot hipIpcOpenMemHandle its own
 # handle, so handles deserialized by the producing process resolve through
-# this registry instead. Strong refs (the producer must keep the storage
-# alive for consumers regardless); a TTL bounds never-consumed entries.
-_local = {}  # key -> (tensor, deadline)
+# this registry instead. reduce_tensor() pre-increments a producer-side IPC
+# ref counter that only a consumer's rebuild/release would decrement, so the
+# same-process path must release it explicitly or every share pins its
+# storage forever (torch reductions.py does the same on its storage-cache
+# hit path). A TTL bounds never-consumed entries.
+_local = {}  # key -> (tensor, deadline, func, args)
 _counter = [0]
 _TTL = 120.0
 
 
+def _release_producer_ref(func, args):
+    if getattr(func, "__name__", "") != "rebuild_cuda_tensor" or len(args) < 13:
+        return
+    storage_cls, device = args[4], args[6]
+    ref_counter_handle, ref_counter_offset = args[11], args[12]
+    try:
+        storage_cls._release_ipc_counter(
+            ref_counter_handle, ref_counter_offset, device=device
+        )
+    except (AttributeError, RuntimeError):
+        pass
+
+
 def _evict():
     now = time.monotonic()
-    dead = [k for k, (_, dl) in _local.items() if dl < now]
+    dead = [k for k, ent in _local.items() if ent[1] < now]
     for k in dead:
-        del _local[k]
+        ent = _local.pop(k)
+        _release_producer_ref(ent[2], ent[3])
 
 
 def _materialize(pid, key, func, args):
@@ -48,6 +65,7 @@ def _materialize(pid, key, func, args):
         # batches in-process doesn't pin them all for the TTL.
         ent = _local.pop(key, None)
         if ent is not None:
+            _release_producer_ref(func, args)
             return ent[0]
     return func(*args)
 
@@ -65,7 +83,7 @@ class SharedCudaTensor:
         _evict()
         _counter[0] += 1
         key = _counter[0]
-        _local[key] = (self._tensor, time.monotonic() + _TTL)
+        _local[key] = (self._tensor, time.monotonic() + _TTL, func, args)
         return (_materialize, (os.getpid(), key, func, args))
 
     def tensor(self):
