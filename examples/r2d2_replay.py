@@ -13,7 +13,6 @@ import time
 import torch
 
 import moolib_amd
-from moolib_amd import ipc
 from moolib_amd.envs import SyntheticAtariEnv
 from moolib_amd.models.atari import AtariNet
 from moolib_amd.replay import ReplayBuffer
@@ -37,13 +36,7 @@ def main():
     server_rpc.set_name("replay_server")
     addr = [a for a in server_rpc.listen("127.0.0.1:0") if a.startswith("tcp://127")][0]
     buf = ReplayBuffer(args.capacity, device=device, alpha=0.6, beta=0.4)
-    buf.serve(server_rpc, "replay")
-    if args.ipc and device.startswith("cuda"):
-        def sample_ipc(batch_size):
-            batch, idx, w = buf.sample(batch_size)
-            return (nest.map(ipc.share, batch), idx.cpu(), w.cpu())
-
-        server_rpc.define("replay.sample_ipc", sample_ipc)
+    buf.serve(server_rpc, "replay")  # serves .sample and .sample_ipc
 
     # Actor: generate LSTM rollouts from synthetic frames.
     envs = moolib_amd.EnvPool(

@@ -12,6 +12,7 @@ importance weights w_i = (N * P(i))^-beta / max w.
 """
 import torch
 
+from moolib_amd import ipc
 from moolib_amd.utils import nest
 
 
@@ -88,13 +89,22 @@ class ReplayBuffer:
         def sample(batch_size):
             batch, idx, w = self.sample(batch_size)
             # Tensors cross the wire on the CPU; receivers move them where
-            # they want. (Same-node GPU->GPU via hipIpc is a planned
-            # upgrade of the transport.)
+            # they want. Same-node consumers use `.sample_ipc` instead.
             return (
                 nest.map(lambda t: t.cpu(), batch),
                 idx.cpu(),
                 w.cpu(),
             )
+
+        def sample_ipc(batch_size):
+            # Zero-copy for same-node consumers: device tensors ship as
+            # hipIpc handles (moolib_amd.ipc); idx/weights are tiny, CPU.
+            batch, idx, w = self.sample(batch_size)
+            if self.device.type == "cuda":
+                batch = nest.map(ipc.share, batch)
+            else:
+                batch = nest.map(lambda t: t.cpu(), batch)
+            return (batch, idx.cpu(), w.cpu())
 
         def update_priorities(indices, priorities):
             self.update_priorities(indices, priorities)
@@ -104,6 +114,7 @@ class ReplayBuffer:
 
         rpc.define(name + ".add", add)
         rpc.define(name + ".sample", sample)
+        rpc.define(name + ".sample_ipc", sample_ipc)
         rpc.define(name + ".update_priorities", update_priorities)
         rpc.define(name + ".info", info)
         return self
@@ -122,6 +133,11 @@ class ReplayClient:
 
     def sample(self, batch_size):
         return self.rpc.async_(self.server, self.name + ".sample", batch_size)
+
+    def sample_ipc(self, batch_size):
+        """Same-node zero-copy sample: device tensors arrive as hipIpc
+        aliases of the server's HBM (do not use across nodes)."""
+        return self.rpc.async_(self.server, self.name + ".sample_ipc", batch_size)
 
     def update_priorities(self, indices, priorities):
         return self.rpc.async_(

@@ -74,3 +74,40 @@ def test_hipipc_tensor_rpc(tmp_path):
         proc.join(timeout=30)
         if proc.is_alive():
             proc.kill()
+
+
+@gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+@pytest.mark.timeout(120)
+def test_hipipc_same_process():
+    """Handles deserialized by the producing process resolve to the original
+    tensor (a process cannot hipIpcOpenMemHandle its own handle), and the
+    producer-side IPC ref counter is released so streams of shares don't
+    pin HBM (moolib_amd/ipc.py registry)."""
+    import moolib_amd
+    from moolib_amd import ipc
+
+    host = moolib_amd.Rpc()
+    host.set_name("same_proc_host")
+    addr = host.listen("127.0.0.1:0")[0]
+    payload = torch.arange(256, dtype=torch.float32, device="cuda")
+    host.define("get", lambda: ipc.share(payload))
+
+    client = moolib_amd.Rpc()
+    client.set_name("same_proc_client")
+    client.set_timeout(60)
+    client.connect(addr)
+
+    t = client.sync("same_proc_host", "get")
+    assert t.is_cuda and t.data_ptr() == payload.data_ptr()
+
+    # a stream of fresh shares must not accumulate pinned storage
+    free0, _ = torch.cuda.mem_get_info()
+    for _ in range(50):
+        host_side = torch.randn(1024 * 1024, device="cuda")  # 4 MiB each
+        host.define("get_fresh", lambda ref=host_side: ipc.share(ref))
+        got = client.sync("same_proc_host", "get_fresh")
+        del host_side, got
+    torch.cuda.synchronize()
+    free1, _ = torch.cuda.mem_get_info()
+    assert free0 - free1 < 64 * 1024 * 1024, (free0 - free1) / 1e6

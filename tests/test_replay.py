@@ -66,3 +66,22 @@ class TestReplayServed:
         # priorities update across the wire
         client.update_priorities(idx, torch.ones(6) * 3.0).result()
         assert len(buf) == 10
+
+    def test_sample_ipc_endpoint_cpu(self):
+        # On a CPU buffer .sample_ipc degrades to the .cpu() wire path, so
+        # clients can use one endpoint everywhere.
+        server = moolib_amd.Rpc()
+        server.set_name("replay_server2")
+        addr = server.listen("127.0.0.1:0")[0]
+        buf = ReplayBuffer(capacity=8).serve(server, "replay")
+
+        client_rpc = moolib_amd.Rpc()
+        client_rpc.set_name("learner2")
+        client_rpc.set_timeout(15)
+        client_rpc.connect(addr)
+        client = ReplayClient(client_rpc, "replay_server2", "replay")
+        for i in range(4):
+            client.add({"obs": torch.full((3,), float(i))}).result()
+        batch, idx, w = client.sample_ipc(5).result()
+        assert batch["obs"].shape == (5, 3)
+        assert not batch["obs"].is_cuda
