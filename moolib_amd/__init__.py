@@ -10,6 +10,7 @@ IMPALA path is hand-written HIP/CDNA4 kernels (moolib_amd.ops).
 import asyncio
 import atexit
 import threading
+import time
 
 __version__ = "0.1.0"
 
@@ -110,28 +111,68 @@ class _BatchCollector:
     the first call; returned tensors are split back per caller.
     """
 
-    def __init__(self, batch_size, device, process, dynamic=False, max_latency=0.01):
+    def __init__(
+        self,
+        batch_size,
+        device,
+        process,
+        dynamic=False,
+        max_latency=0.01,
+        min_latency=0.0005,
+    ):
         self.batch_size = batch_size
         self.device = device
         self.process = process  # fn(batched_args, batched_kwargs, respond_all)
         self.dynamic = dynamic
         self.max_latency = max_latency
+        self.min_latency = min_latency
         self.lock = threading.Lock()
         self.pending = []
         self.timer = None
+        # Latency model (reference src/moolib.cc dynamic batching): the
+        # collection window tracks the measured batch service time — while
+        # one batch computes, the next one fills, so waiting ~proc_ema adds
+        # throughput without adding pipeline latency. arrival_ema estimates
+        # whether a partial batch is worth holding for more arrivals.
+        self.proc_ema = None  # EMA of process() wall time
+        self.arrival_ema = None  # EMA of inter-arrival gap
+        self._last_arrival = None
+
+    def _window(self):
+        if not self.dynamic or self.proc_ema is None:
+            return self.max_latency
+        return min(max(self.proc_ema, self.min_latency), self.max_latency)
 
     def add(self, deferred, args, kwargs):
         fire = None
+        now = time.monotonic()
         with self.lock:
+            if self._last_arrival is not None:
+                gap = now - self._last_arrival
+                self.arrival_ema = (
+                    gap
+                    if self.arrival_ema is None
+                    else 0.8 * self.arrival_ema + 0.2 * gap
+                )
+            self._last_arrival = now
             self.pending.append((deferred, args, kwargs))
-            if len(self.pending) >= self.batch_size:
+            n = len(self.pending)
+            full = n >= self.batch_size
+            # Early flush: if filling the rest of the batch is expected to
+            # take longer than the window, waiting only adds latency.
+            stale = (
+                self.dynamic
+                and self.arrival_ema is not None
+                and (self.batch_size - n) * self.arrival_ema > self._window()
+            )
+            if full or stale:
                 fire = self.pending
                 self.pending = []
                 if self.timer is not None:
                     self.timer.cancel()
                     self.timer = None
             elif self.dynamic and self.timer is None:
-                self.timer = threading.Timer(self.max_latency, self._flush)
+                self.timer = threading.Timer(self._window(), self._flush)
                 self.timer.daemon = True
                 self.timer.start()
         if fire:
@@ -169,7 +210,11 @@ class _BatchCollector:
                 )
                 deferred(out_i)
 
+        t0 = time.monotonic()
         self.process(tuple(batched_args), dict(batched_kwargs), respond_all, n)
+        dt = time.monotonic() - t0
+        with self.lock:
+            self.proc_ema = dt if self.proc_ema is None else 0.8 * self.proc_ema + 0.2 * dt
 
 
 class Rpc(_core.Rpc):

@@ -256,3 +256,53 @@ class TestDynamicBatching:
         # partial batches flushed by the latency timer (arrival timing may
         # split them, but nothing waits for a full batch of 8)
         assert sum(seen_sizes) == 3 and max(seen_sizes) < 8, seen_sizes
+
+
+class TestDynamicBatchingLatencyModel:
+    """The collector adapts: full batches fire at once, slow arrivals flush
+    early (holding a partial batch longer than filling it would take only
+    adds latency), and the window tracks measured service time."""
+
+    def _collector(self, batch_size, fired, proc_time=0.0):
+        import moolib_amd as M
+
+        def process(bargs, bkwargs, respond_all, n):
+            if proc_time:
+                time.sleep(proc_time)
+            fired.append((n, time.monotonic()))
+
+        return M._BatchCollector(batch_size, None, process, dynamic=True)
+
+    def test_full_batch_fires_immediately(self):
+        fired = []
+        c = self._collector(4, fired)
+        t0 = time.monotonic()
+        for _ in range(4):
+            c.add(lambda r: None, (), {})
+        assert len(fired) == 1 and fired[0][0] == 4
+        assert fired[0][1] - t0 < 0.01
+
+    def test_slow_arrivals_flush_early(self):
+        fired = []
+        c = self._collector(8, fired)
+        lat = []
+        for i in range(5):
+            t0 = time.monotonic()
+            c.add(lambda r: None, (), {})
+            # give the first (timer-driven) flush a chance, later ones are
+            # early-flushed inline by the arrival model
+            time.sleep(0.05)
+            if fired:
+                lat.append(fired[-1][1] - t0)
+        assert len(fired) >= 4, fired
+        # calls 2+ fire inline (arrival gap 50ms >> window): near-zero wait
+        assert sorted(lat)[len(lat) // 2] < 0.02, lat
+
+    def test_window_tracks_service_time(self):
+        fired = []
+        c = self._collector(64, fired, proc_time=0.005)
+        c.add(lambda r: None, (), {})
+        c._flush()
+        assert c.proc_ema is not None and c.proc_ema >= 0.004
+        w = c._window()
+        assert 0.004 <= w <= c.max_latency
