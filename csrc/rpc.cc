@@ -200,9 +200,18 @@ void Rpc::onClosed(ConnId id, const std::string& reason) {
   std::lock_guard<std::mutex> lk(mu_);
   auto it = conns_.find(id);
   std::string peerName;
+  std::string connAddr;
+  bool wasReady = false;
   if (it != conns_.end()) {
     peerName = it->second.peerName;
+    connAddr = it->second.addr;
+    wasReady = it->second.ready;
     conns_.erase(it);
+  }
+  if (!connAddr.empty() && !wasReady && !peerName.empty()) {
+    // dial failed before greeting: this transport is suspect
+    auto pit = peers_.find(peerName);
+    if (pit != peers_.end()) pit->second.transport[connAddr].failPenalty += 1.0;
   }
   if (!peerName.empty()) {
     auto pit = peers_.find(peerName);
@@ -278,23 +287,30 @@ void Rpc::tryConnectPeerLocked(const std::string& name, PeerInfo& p) {
   if (p.activeConn != 0 || p.connecting != 0 || p.addrs.empty()) return;
   if (secondsSince(p.lastConnectAttempt) < 0.25) return;
   p.lastConnectAttempt = now();
-  // Prefer unix (same machine) addresses.
+  // Latency-informed transport choice: score every candidate address by its
+  // measured round-trip EMA plus a decaying failure penalty; unexplored
+  // addresses get optimistic priors (unix beats tcp on the same machine
+  // until data says otherwise). The resend-on-reconnect + receiver-dedupe
+  // machinery makes switching transports between attempts safe.
   std::string addr;
-  for (size_t i = 0; i < p.addrs.size(); ++i) {
-    const std::string& cand = p.addrs[(p.nextAddr + i) % p.addrs.size()];
-    if (cand.rfind("unix://", 0) == 0) {
+  double best = 0;
+  for (auto& cand : p.addrs) {
+    TransportStat& st = p.transport[cand];
+    st.failPenalty *= 0.5;  // dead transports get retried eventually
+    double score = st.samples == 0
+                       ? (cand.rfind("unix://", 0) == 0 ? -2.0 : -1.0)
+                       : st.ema;
+    score += st.failPenalty;
+    if (addr.empty() || score < best) {
+      best = score;
       addr = cand;
-      p.nextAddr = (p.nextAddr + i + 1) % p.addrs.size();
-      break;
     }
   }
-  if (addr.empty()) {
-    addr = p.addrs[p.nextAddr % p.addrs.size()];
-    p.nextAddr = (p.nextAddr + 1) % p.addrs.size();
-  }
+  if (addr.empty()) return;
   p.connecting = engine_->connect(addr);
   ConnInfo ci;
   ci.peerName = name;
+  ci.addr = addr;
   ci.established = now();
   ci.lastRecv = now();
   conns_[p.connecting] = ci;
@@ -509,6 +525,12 @@ void Rpc::handleResponse(ConnId id, Frame&& f, bool isError) {
     double lat = secondsSince(it->second.sentAt);
     PeerInfo& p = getPeer(peerName);
     p.latencyEma = p.latencyEma == 0 ? lat : p.latencyEma * 0.9 + lat * 0.1;
+    auto cit = conns_.find(it->second.sentOn);
+    if (cit != conns_.end() && !cit->second.addr.empty()) {
+      TransportStat& st = p.transport[cit->second.addr];
+      st.ema = st.ema < 0 ? lat : st.ema * 0.9 + lat * 0.1;
+      ++st.samples;
+    }
     ++p.recvCount;
     uint64_t fbytes = f.payload.size();
     for (auto& t : f.tensors) fbytes += t.nbytes();
@@ -662,7 +684,12 @@ std::string Rpc::debugInfo() {
        << p.recvCount << " bytes tx/rx=" << p.bytesSent << "/" << p.bytesRecv
        << " latency_ema=" << p.latencyEma * 1000 << "ms addrs=[";
     for (auto& a : p.addrs) os << a << ",";
-    os << "]\n";
+    os << "] transport={";
+    for (auto& [a, st] : p.transport) {
+      os << a << ": ema=" << st.ema * 1000 << "ms n=" << st.samples
+         << " fail=" << st.failPenalty << ", ";
+    }
+    os << "}\n";
   }
   os << "  outgoing in flight: " << outgoing_.size() << "\n";
   os << "  incoming tracked: " << incoming_.size() << "\n";

@@ -18,6 +18,7 @@
 #include <atomic>
 #include <deque>
 #include <memory>
+#include <map>
 #include <unordered_map>
 #include <unordered_set>
 
@@ -79,10 +80,21 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
   struct ConnInfo {
     std::string peerUid;   // empty until greeting received
     std::string peerName;
+    std::string addr;      // dial address (outbound only; empty if inbound)
     bool ready = false;
     bool inbound = false;
     TimePoint lastRecv{};
     TimePoint established{};
+  };
+
+  // Per-address transport model (reference rpc.cc:640 "bandit" transport
+  // choice, MI355X-shaped: round-trip latency EMA per dial address plus a
+  // decaying failure penalty; reconnects pick the argmin, unexplored
+  // addresses get optimistic priors so every transport gets sampled).
+  struct TransportStat {
+    double ema = -1;          // seconds; <0 = never sampled
+    uint64_t samples = 0;
+    double failPenalty = 0;   // added to score, halves per connect cycle
   };
 
   struct PeerInfo {
@@ -98,6 +110,7 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
     uint64_t bytesSent = 0;
     uint64_t bytesRecv = 0;
     double latencyEma = 0;  // seconds, over request->response round trips
+    std::map<std::string, TransportStat> transport;  // keyed by dial addr
   };
 
   struct Outgoing {

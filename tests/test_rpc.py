@@ -306,3 +306,22 @@ class TestDynamicBatchingLatencyModel:
         assert c.proc_ema is not None and c.proc_ema >= 0.004
         w = c._window()
         assert 0.004 <= w <= c.max_latency
+
+
+class TestTransportModel:
+    def test_latency_attributed_to_transport(self):
+        host = moolib_amd.Rpc()
+        host.set_name("tm_host")
+        addr = host.listen("127.0.0.1:0")[0]
+        client = moolib_amd.Rpc()
+        client.set_name("tm_client")
+        client.set_timeout(15)
+        client.connect(addr)
+        host.define("ping", lambda: "pong")
+        for _ in range(5):
+            assert client.sync("tm_host", "ping") == "pong"
+        info = client.debug_info()
+        assert "transport={" in info
+        # the dialed tcp address has samples and a measured ema
+        line = [l for l in info.splitlines() if "tm_host" in l and "transport=" in l][0]
+        assert "ema=" in line and "n=5" in line, line
