@@ -118,6 +118,7 @@ void Rpc::connect(const std::string& addr) {
   e.lastAttempt = now();
   e.conn = engine_->connect(addr);
   ConnInfo ci;
+  ci.addr = addr;
   ci.established = now();
   ci.lastRecv = now();
   conns_[e.conn] = ci;
@@ -618,6 +619,7 @@ void Rpc::timerLoop() {
           e.backoff = std::min(e.backoff * 1.6, 2.0);
           e.conn = engine_->connect(e.addr);
           ConnInfo ci;
+          ci.addr = e.addr;
           ci.established = now();
           ci.lastRecv = now();
           conns_[e.conn] = ci;
