@@ -12,10 +12,12 @@
 #include <sys/socket.h>
 #include <sys/uio.h>
 #include <sys/un.h>
+#include <pthread.h>
 #include <unistd.h>
 
 #include <cstring>
 #include <deque>
+#include <set>
 #include <unordered_map>
 
 namespace mrl {
@@ -136,10 +138,56 @@ struct Conn {
   size_t frontOffset = 0;  // bytes of writeQ.front() already sent
 };
 
+// Fork hygiene: EnvPool forks worker processes that never exec, so every
+// engine fd (sockets, epoll, eventfd) would stay open in the children and
+// keep ports/peer connections alive after the parent dies. A pthread_atfork
+// child handler closes all registered engine fds. The fd set has its own
+// mutex (locked across the fork) because the conn maps are epoll-thread-
+// owned and may be mid-mutation at fork time.
+namespace {
+std::mutex* forkRegMutex() {
+  static std::mutex* m = new std::mutex();  // leaked: outlives exit order
+  return m;
+}
+std::set<std::set<int>*>* forkRegSet() {
+  static std::set<std::set<int>*>* s = new std::set<std::set<int>*>();
+  return s;
+}
+void forkPrepare() { forkRegMutex()->lock(); }
+void forkParent() { forkRegMutex()->unlock(); }
+void forkChild() {
+  for (std::set<int>* fds : *forkRegSet()) {
+    for (int fd : *fds) ::close(fd);
+    fds->clear();
+  }
+  forkRegMutex()->unlock();
+}
+void registerForkHygiene(std::set<int>* fds) {
+  static std::once_flag once;
+  std::call_once(once, [] { pthread_atfork(forkPrepare, forkParent, forkChild); });
+  std::lock_guard<std::mutex> lk(*forkRegMutex());
+  forkRegSet()->insert(fds);
+}
+void unregisterForkHygiene(std::set<int>* fds) {
+  std::lock_guard<std::mutex> lk(*forkRegMutex());
+  forkRegSet()->erase(fds);
+}
+}  // namespace
+
 struct SocketEngine::Impl {
   SocketEngineCallbacks cbs;
   int epfd = -1;
   int wakeFd = -1;
+  std::set<int> ownedFds;  // guarded by the fork-hygiene mutex
+
+  void trackFd(int fd) {
+    std::lock_guard<std::mutex> lk(*forkRegMutex());
+    ownedFds.insert(fd);
+  }
+  void untrackFd(int fd) {
+    std::lock_guard<std::mutex> lk(*forkRegMutex());
+    ownedFds.erase(fd);
+  }
   std::thread thread;
   std::atomic<bool> stopping{false};
   std::atomic<uint64_t> nextId{1};
@@ -182,6 +230,7 @@ struct SocketEngine::Impl {
     int fd = it->second.fd;
     epollCtl(EPOLL_CTL_DEL, fd, 0, 0);
     ::close(fd);
+    untrackFd(fd);
     fdToConn.erase(fd);
     conns.erase(it);
     if (cbs.onClosed && !stopping.load()) cbs.onClosed(id, reason);
@@ -374,6 +423,7 @@ struct SocketEngine::Impl {
       }
       if (!isUnix) setTcpOpts(fd);
       setBufSizes(fd);
+      trackFd(fd);
       ConnId id = nextId.fetch_add(1);
       Conn& c = conns[id];
       c.fd = fd;
@@ -457,6 +507,9 @@ SocketEngine::SocketEngine(SocketEngineCallbacks cbs) : impl_(new Impl()) {
   impl_->wakeFd = eventfd(0, EFD_NONBLOCK | EFD_CLOEXEC);
   if (impl_->epfd < 0 || impl_->wakeFd < 0) throw RpcError("epoll/eventfd creation failed");
   impl_->epollCtl(EPOLL_CTL_ADD, impl_->wakeFd, EPOLLIN, 0);
+  impl_->trackFd(impl_->epfd);
+  impl_->trackFd(impl_->wakeFd);
+  registerForkHygiene(&impl_->ownedFds);
   impl_->thread = std::thread([this] { impl_->loop(); });
 }
 
@@ -477,6 +530,11 @@ void SocketEngine::shutdown() {
   impl_->listeners.clear();
   ::close(impl_->epfd);
   ::close(impl_->wakeFd);
+  unregisterForkHygiene(&impl_->ownedFds);
+  {
+    std::lock_guard<std::mutex> lk(*forkRegMutex());
+    impl_->ownedFds.clear();
+  }
 }
 
 ConnId SocketEngine::connect(const std::string& addrStr) {
@@ -522,6 +580,7 @@ ConnId SocketEngine::connect(const std::string& addrStr) {
       if (impl_->cbs.onClosed) impl_->cbs.onClosed(id, reason);
       return;
     }
+    impl_->trackFd(fd);
     Conn& c = impl_->conns[id];
     c.fd = fd;
     c.isUnix = isUnix;
@@ -582,6 +641,7 @@ std::vector<std::string> SocketEngine::listen(const std::string& addrStr) {
     throw RpcError(std::string("listen failed: ") + strerror(errno));
   }
   bool isUnix = a.scheme == "unix";
+  impl_->trackFd(fd);
   impl_->post([this, fd, isUnix] {
     impl_->listeners[fd] = isUnix;
     impl_->epollCtl(EPOLL_CTL_ADD, fd, EPOLLIN, (uint64_t(1) << 63) | static_cast<uint32_t>(fd));

@@ -116,3 +116,41 @@ class TestEnvPool:
         pool = moolib_amd.EnvPool(CountingEnv, num_processes=2, batch_size=2, num_batches=1)
         assert pool.running()
         assert pool.num_workers_alive() == 2
+
+
+class FdProbeEnv:
+    """Observation = number of open socket fds in the worker process."""
+
+    def reset(self):
+        return self._probe()
+
+    def step(self, action):
+        return self._probe(), 0.0, False, {}
+
+    def _probe(self):
+        import os
+
+        n = 0
+        for fd in os.listdir("/proc/self/fd"):
+            try:
+                if "socket:" in os.readlink("/proc/self/fd/" + fd):
+                    n += 1
+            except OSError:
+                pass
+        return np.array([n], dtype=np.int64)
+
+
+class TestForkHygiene:
+    def test_workers_inherit_no_engine_sockets(self):
+        # A live RPC plane in the parent must not leak its sockets into
+        # EnvPool workers (pthread_atfork handler in csrc/socket.cc).
+        rpc = moolib_amd.Rpc()
+        rpc.set_name("fd_parent")
+        rpc.listen("127.0.0.1:0")
+        pool = moolib_amd.EnvPool(FdProbeEnv, num_processes=2, batch_size=2, num_batches=1)
+        obs = pool.step(0, torch.zeros(2, dtype=torch.int64)).result()
+        import moolib_amd.utils.nest as nest
+
+        counts = list(nest.flatten(obs))[0]
+        assert int(counts.max()) == 0, counts
+        del pool
