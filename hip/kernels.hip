@@ -528,11 +528,169 @@ at::Tensor maxpool3x3s2_bwd(at::Tensor gout, at::Tensor idx, int64_t H, int64_t 
   return gin;
 }
 
+// ------------------------- fused conv-bias epilogues -----------------------
+// MIOpen applies conv bias as its own full-tensor pass (SubTensorOpWithScalar
+// in the kernel trace) and torch's relu / residual add are two more passes
+// over HBM. Running the convs bias-free (F.conv2d(..., None)) and folding the
+// bias into the next elementwise op turns (bias, relu) and (bias, add) into
+// ONE bandwidth-bound pass each. Per-channel bias commutes with the
+// per-channel spatial maxpool, so a section conv's bias rides through the
+// pool into the next block's bias_relu and the residual shortcut (the second
+// bias of bias_add2). Model integration: moolib_amd/models/atari.py
+// (reference architecture examples/atari/models.py:9-153 is unchanged
+// mathematically).
+
+template <typename T>
+__global__ void bias_relu_fwd_kernel(const T* __restrict__ x, const T* __restrict__ b,
+                                     T* __restrict__ y, int64_t total8, int c8n) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= total8) return;
+  int c8 = tid % c8n;
+  const Vec8<T> vx = *reinterpret_cast<const Vec8<T>*>(x + tid * 8);
+  const Vec8<T> vb = *reinterpret_cast<const Vec8<T>*>(b + (int64_t)c8 * 8);
+  Vec8<T> vy;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    float v = (float)vx.v[j] + (float)vb.v[j];
+    vy.v[j] = (T)(v > 0.f ? v : 0.f);
+  }
+  *reinterpret_cast<Vec8<T>*>(y + tid * 8) = vy;
+}
+
+// dx = dy * (y > 0); db accumulated per-workgroup in LDS then one global
+// atomic per channel (C <= 64 in this model family).
+template <typename T>
+__global__ void bias_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ y,
+                                     T* __restrict__ dx, float* __restrict__ db,
+                                     int64_t total8, int c8n) {
+  __shared__ float lds_db[64];
+  int C = c8n * 8;
+  if (threadIdx.x < C) lds_db[threadIdx.x] = 0.f;
+  __syncthreads();
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid < total8) {
+    int c8 = tid % c8n;
+    const Vec8<T> vdy = *reinterpret_cast<const Vec8<T>*>(dy + tid * 8);
+    const Vec8<T> vy = *reinterpret_cast<const Vec8<T>*>(y + tid * 8);
+    Vec8<T> vdx;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float g = (float)vy.v[j] > 0.f ? (float)vdy.v[j] : 0.f;
+      vdx.v[j] = (T)g;
+      atomicAdd(&lds_db[c8 * 8 + j], g);
+    }
+    *reinterpret_cast<Vec8<T>*>(dx + tid * 8) = vdx;
+  }
+  __syncthreads();
+  if (threadIdx.x < C) atomicAdd(&db[threadIdx.x], lds_db[threadIdx.x]);
+}
+
+// y = x + b1 + s (+ b2): residual close with the producing conv's bias and,
+// when the shortcut came through a pooled bias-free section conv, that
+// conv's pending bias too.
+template <typename T>
+__global__ void bias_add2_fwd_kernel(const T* __restrict__ x, const T* __restrict__ b1,
+                                     const T* __restrict__ s, const T* __restrict__ b2,
+                                     T* __restrict__ y, int64_t total8, int c8n) {
+  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (tid >= total8) return;
+  int c8 = tid % c8n;
+  const Vec8<T> vx = *reinterpret_cast<const Vec8<T>*>(x + tid * 8);
+  const Vec8<T> vs = *reinterpret_cast<const Vec8<T>*>(s + tid * 8);
+  const Vec8<T> vb1 = *reinterpret_cast<const Vec8<T>*>(b1 + (int64_t)c8 * 8);
+  Vec8<T> vy;
+  if (b2 != nullptr) {
+    const Vec8<T> vb2 = *reinterpret_cast<const Vec8<T>*>(b2 + (int64_t)c8 * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      vy.v[j] = (T)((float)vx.v[j] + (float)vb1.v[j] + (float)vs.v[j] + (float)vb2.v[j]);
+  } else {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      vy.v[j] = (T)((float)vx.v[j] + (float)vb1.v[j] + (float)vs.v[j]);
+  }
+  *reinterpret_cast<Vec8<T>*>(y + tid * 8) = vy;
+}
+
+static void checkNhwcPair(const at::Tensor& x, const char* who) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4, who, ": 4D CUDA tensor expected");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), who, ": channels_last expected");
+  TORCH_CHECK(x.size(1) % 8 == 0, who, ": channels must be a multiple of 8");
+}
+
+at::Tensor bias_relu_fwd(at::Tensor x, at::Tensor b) {
+  checkNhwcPair(x, "bias_relu");
+  int C = x.size(1);
+  TORCH_CHECK(b.numel() == C && b.scalar_type() == x.scalar_type(), "bias_relu: bad bias");
+  auto out = at::empty_like(x, x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total8 = x.numel() / 8;
+  int threads = 256;
+  int64_t blocks = (total8 + threads - 1) / threads;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, x.scalar_type(), "bias_relu_fwd", [&] {
+        hipLaunchKernelGGL(bias_relu_fwd_kernel<scalar_t>, dim3(blocks), dim3(threads), 0, stream,
+                           x.data_ptr<scalar_t>(), b.contiguous().data_ptr<scalar_t>(),
+                           out.data_ptr<scalar_t>(), total8, C / 8);
+      });
+  return out;
+}
+
+std::vector<at::Tensor> bias_relu_bwd(at::Tensor dy, at::Tensor y) {
+  checkNhwcPair(y, "bias_relu_bwd");
+  auto g = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  int C = y.size(1);
+  TORCH_CHECK(C <= 64, "bias_relu_bwd: C <= 64 (LDS accumulator)");
+  auto dx = at::empty_like(y, y.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto db = at::zeros({C}, y.options().dtype(at::kFloat));
+  int64_t total8 = y.numel() / 8;
+  int threads = 256;
+  int64_t blocks = (total8 + threads - 1) / threads;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, y.scalar_type(), "bias_relu_bwd", [&] {
+        hipLaunchKernelGGL(bias_relu_bwd_kernel<scalar_t>, dim3(blocks), dim3(threads), 0, stream,
+                           g.data_ptr<scalar_t>(), y.data_ptr<scalar_t>(),
+                           dx.data_ptr<scalar_t>(), db.data_ptr<float>(), total8, C / 8);
+      });
+  return {dx, db};
+}
+
+at::Tensor bias_add2_fwd(at::Tensor x, at::Tensor b1, at::Tensor s,
+                         c10::optional<at::Tensor> b2) {
+  checkNhwcPair(x, "bias_add2");
+  checkNhwcPair(s, "bias_add2(shortcut)");
+  int C = x.size(1);
+  TORCH_CHECK(b1.numel() == C && b1.scalar_type() == x.scalar_type(), "bias_add2: bad bias1");
+  at::Tensor b2c;
+  if (b2.has_value()) {
+    TORCH_CHECK(b2->numel() == C && b2->scalar_type() == x.scalar_type(), "bias_add2: bad bias2");
+    b2c = b2->contiguous();
+  }
+  auto out = at::empty_like(x, x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  int64_t total8 = x.numel() / 8;
+  int threads = 256;
+  int64_t blocks = (total8 + threads - 1) / threads;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, x.scalar_type(), "bias_add2_fwd", [&] {
+        hipLaunchKernelGGL(bias_add2_fwd_kernel<scalar_t>, dim3(blocks), dim3(threads), 0, stream,
+                           x.data_ptr<scalar_t>(), b1.contiguous().data_ptr<scalar_t>(),
+                           s.data_ptr<scalar_t>(),
+                           b2.has_value() ? b2c.data_ptr<scalar_t>() : nullptr,
+                           out.data_ptr<scalar_t>(), total8, C / 8);
+      });
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "moolib_amd gfx950 HIP kernels";
   m.def("register_host_memory", &register_host_memory);
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd, "NHWC 3x3/2 maxpool forward (gfx950)");
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd, "NHWC 3x3/2 maxpool backward (gather, no atomics)");
+  m.def("bias_relu_fwd", &bias_relu_fwd, "fused conv-bias + relu, one NHWC pass (gfx950)");
+  m.def("bias_relu_bwd", &bias_relu_bwd, "bias_relu backward: dx + fp32 db in one pass");
+  m.def("bias_add2_fwd", &bias_add2_fwd, "fused residual close: x + bias1 + shortcut (+ bias2)");
   m.def("frames_u8_to_bf16_nhwc", &frames_u8_to_bf16_nhwc, "fused uint8->bf16 NHWC scale");
   m.def("conv1_u8_nhwc", &conv1_u8_nhwc, "fused uint8 frames -> conv(4->16,3x3)+bias, NHWC bf16");
   m.def("lstm_fused_fwd", &lstm_fused_fwd, "fused masked LSTM sequence scan fwd (MFMA, gfx950)");

@@ -28,6 +28,23 @@ class ResidualBlock(nn.Module):
         y = self.conv1(F.relu(y))
         return x + y
 
+    def forward_fused(self, x, pending_bias=None):
+        """Same math with the conv biases folded into the elementwise ops
+        (ops/fused_bias): convs run bias-free, relu/add are one pass each.
+        `pending_bias` is a per-channel bias the CALLER still owes `x`
+        (section conv bias carried through the maxpool).
+        """
+        from moolib_amd.ops.fused_bias import bias_add2, bias_relu
+
+        if pending_bias is not None:
+            t = bias_relu(x, pending_bias)
+        else:
+            t = F.relu(x)
+        u = F.conv2d(t, self.conv0.weight, None, padding=1)
+        t = bias_relu(u, self.conv0.bias)
+        u = F.conv2d(t, self.conv1.weight, None, padding=1)
+        return bias_add2(u, self.conv1.bias, x, pending_bias)
+
 
 class ConvSection(nn.Module):
     def __init__(self, in_ch, out_ch):
@@ -38,8 +55,16 @@ class ConvSection(nn.Module):
         self.res1 = ResidualBlock(out_ch)
 
     def forward(self, x):
+        from moolib_amd.ops import fused_bias
         from moolib_amd.ops.pool import maxpool3x3s2
 
+        if fused_bias.available(x, self.conv.out_channels):
+            # bias-free conv; per-channel bias commutes with the per-channel
+            # spatial max, so it rides into res0 as a pending bias.
+            u = F.conv2d(x, self.conv.weight, None, padding=1)
+            u = maxpool3x3s2(u)
+            u = self.res0.forward_fused(u, self.conv.bias)
+            return self.res1.forward_fused(u)
         x = maxpool3x3s2(self.conv(x))
         x = self.res0(x)
         x = self.res1(x)
@@ -112,7 +137,13 @@ class AtariNet(nn.Module):
             from moolib_amd.ops.pool import maxpool3x3s2
 
             sec0 = self.sections[0]
-            x = sec0.res1(sec0.res0(maxpool3x3s2(x)))
+            x = maxpool3x3s2(x)
+            from moolib_amd.ops import fused_bias
+
+            if fused_bias.available(x):
+                x = sec0.res1.forward_fused(sec0.res0.forward_fused(x))
+            else:
+                x = sec0.res1(sec0.res0(x))
             first = 1
         elif kernels is not None and not _os.environ.get("MOOLIB_AMD_NO_FRAMES_KERNEL"):
             x = kernels.frames_u8_to_bf16_nhwc(x, 1.0 / 255.0)
