@@ -352,6 +352,8 @@ class TestConv1Kernel:
         assert d < 0.25, d  # bf16 accumulation-order differences through the net
 
 
+@gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
 class TestFusedBias:
     """bias_relu / bias_add2 vs plain fp32 torch (exact in fp32)."""
 
