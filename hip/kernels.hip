@@ -557,8 +557,12 @@ __global__ void bias_relu_fwd_kernel(const T* __restrict__ x, const T* __restric
   *reinterpret_cast<Vec8<T>*>(y + tid * 8) = vy;
 }
 
-// dx = dy * (y > 0); db accumulated per-workgroup in LDS then one global
-// atomic per channel (C <= 64 in this model family).
+// dx = dy * (y > 0); db reduced without contention: a grid-stride loop with
+// stride = blockDim*gridDim (a multiple of c8n, so each thread's channel
+// group is FIXED) accumulates db partials in registers, then one LDS pass
+// and one global atomic per channel per workgroup. The naive
+// one-atomic-per-element version serialized 256 threads onto C addresses
+// and ran 14x slower than bandwidth.
 template <typename T>
 __global__ void bias_relu_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ y,
                                      T* __restrict__ dx, float* __restrict__ db,
@@ -567,9 +571,11 @@ __global__ void bias_relu_bwd_kernel(const T* __restrict__ dy, const T* __restri
   int C = c8n * 8;
   if (threadIdx.x < C) lds_db[threadIdx.x] = 0.f;
   __syncthreads();
-  int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (tid < total8) {
-    int c8 = tid % c8n;
+  int64_t stride = (int64_t)blockDim.x * gridDim.x;  // multiple of c8n
+  int64_t tid0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  int c8 = (int)(tid0 % c8n);
+  for (int64_t tid = tid0; tid < total8; tid += stride) {
     const Vec8<T> vdy = *reinterpret_cast<const Vec8<T>*>(dy + tid * 8);
     const Vec8<T> vy = *reinterpret_cast<const Vec8<T>*>(y + tid * 8);
     Vec8<T> vdx;
@@ -577,10 +583,12 @@ __global__ void bias_relu_bwd_kernel(const T* __restrict__ dy, const T* __restri
     for (int j = 0; j < 8; ++j) {
       float g = (float)vy.v[j] > 0.f ? (float)vdy.v[j] : 0.f;
       vdx.v[j] = (T)g;
-      atomicAdd(&lds_db[c8 * 8 + j], g);
+      acc[j] += g;
     }
     *reinterpret_cast<Vec8<T>*>(dx + tid * 8) = vdx;
   }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(&lds_db[c8 * 8 + j], acc[j]);
   __syncthreads();
   if (threadIdx.x < C) atomicAdd(&db[threadIdx.x], lds_db[threadIdx.x]);
 }
@@ -645,7 +653,9 @@ std::vector<at::Tensor> bias_relu_bwd(at::Tensor dy, at::Tensor y) {
   auto db = at::zeros({C}, y.options().dtype(at::kFloat));
   int64_t total8 = y.numel() / 8;
   int threads = 256;
-  int64_t blocks = (total8 + threads - 1) / threads;
+  // grid-stride: enough workgroups to fill 256 CUs across 8 XCDs without
+  // letting the per-block channel atomics dominate
+  int64_t blocks = std::min<int64_t>((total8 + threads - 1) / threads, 2048);
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, y.scalar_type(), "bias_relu_bwd", [&] {

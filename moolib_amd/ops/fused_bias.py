@@ -73,8 +73,7 @@ class _BiasAdd2(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        db = dy.float().sum(dim=(0, 2, 3))
-        db = db.to(ctx.b_dtype)
+        db = dy.sum(dim=(0, 2, 3), dtype=torch.float32).to(ctx.b_dtype)
         return dy, db, dy, (db if ctx.has_b2 else None)
 
 
