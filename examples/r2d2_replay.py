@@ -19,17 +19,17 @@ from moolib_amd.replay import ReplayBuffer
 from moolib_amd.utils import nest
 
 
-def main():
-    ap = argparse.ArgumentParser()
-    ap.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
-    ap.add_argument("--capacity", type=int, default=2048)
-    ap.add_argument("--unroll", type=int, default=40)
-    ap.add_argument("--batch-size", type=int, default=16)
-    ap.add_argument("--num-envs", type=int, default=32)
-    ap.add_argument("--seconds", type=float, default=20.0)
-    ap.add_argument("--ipc", action="store_true", help="serve samples as hipIpc handles")
-    args = ap.parse_args()
-    device = args.device
+def run(device=None, capacity=2048, unroll=40, batch_size=16, num_envs=32,
+        seconds=20.0, ipc=False):
+    """Run the replay workload; returns a metrics dict (also used by
+    `bench.py --config r2d2`)."""
+    device = device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+
+    class args:  # keep the body below unchanged
+        pass
+
+    args.capacity, args.unroll, args.batch_size = capacity, unroll, batch_size
+    args.num_envs, args.seconds, args.ipc = num_envs, seconds, ipc
 
     # Replay server peer: sequences live in HBM.
     server_rpc = moolib_amd.Rpc()
@@ -96,10 +96,40 @@ def main():
             sampled += args.batch_size
 
     dt = time.time() - t0
+    return {
+        "seconds": dt,
+        "frames_acted": frames,
+        "frames_per_s": frames / dt,
+        "buffer_seqs": len(buf),
+        "capacity": args.capacity,
+        "sampled": sampled,
+        "sampled_per_s": sampled / dt,
+        "unroll": args.unroll,
+        "batch_size": args.batch_size,
+        "num_envs": args.num_envs,
+        "ipc": ipc,
+        "device": str(device),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--capacity", type=int, default=2048)
+    ap.add_argument("--unroll", type=int, default=40)
+    ap.add_argument("--batch-size", type=int, default=16)
+    ap.add_argument("--num-envs", type=int, default=32)
+    ap.add_argument("--seconds", type=float, default=20.0)
+    ap.add_argument("--ipc", action="store_true", help="serve samples as hipIpc handles")
+    a = ap.parse_args()
+    m = run(a.device, a.capacity, a.unroll, a.batch_size, a.num_envs, a.seconds, a.ipc)
     print(
         "r2d2 replay demo: %.1fs, %d frames acted (%.0f/s), buffer %d/%d seqs, "
         "%d sequences sampled (%.0f/s)"
-        % (dt, frames, frames / dt, len(buf), args.capacity, sampled, sampled / dt)
+        % (
+            m["seconds"], m["frames_acted"], m["frames_per_s"], m["buffer_seqs"],
+            m["capacity"], m["sampled"], m["sampled_per_s"],
+        )
     )
 
 
