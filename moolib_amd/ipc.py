@@ -21,6 +21,7 @@ Lifetime: the producer's storage stays alive until every consumer drops
 its alias (torch's IPC refcounting). Do not send handles across nodes.
 """
 import os
+import threading
 import time
 
 import torch
@@ -34,6 +35,7 @@ from torch.multiprocessing import reductions
 # storage forever (torch reductions.py does the same on its storage-cache
 # hit path). A TTL bounds never-consumed entries.
 _local = {}  # key -> (tensor, deadline, func, args)
+_lock = threading.Lock()  # shares/materializes run on concurrent RPC threads
 _counter = [0]
 _TTL = 120.0
 
@@ -51,19 +53,19 @@ def _release_producer_ref(func, args):
         pass
 
 
-def _evict():
+def _evict_locked():
     now = time.monotonic()
     dead = [k for k, ent in _local.items() if ent[1] < now]
-    for k in dead:
-        ent = _local.pop(k)
-        _release_producer_ref(ent[2], ent[3])
+    evicted = [_local.pop(k) for k in dead]
+    return evicted
 
 
 def _materialize(pid, key, func, args):
     if pid == os.getpid():
         # Pop: the consumer takes ownership, so serving a stream of fresh
         # batches in-process doesn't pin them all for the TTL.
-        ent = _local.pop(key, None)
+        with _lock:
+            ent = _local.pop(key, None)
         if ent is not None:
             _release_producer_ref(func, args)
             return ent[0]
@@ -80,10 +82,13 @@ class SharedCudaTensor:
 
     def __reduce__(self):
         func, args = reductions.reduce_tensor(self._tensor)
-        _evict()
-        _counter[0] += 1
-        key = _counter[0]
-        _local[key] = (self._tensor, time.monotonic() + _TTL, func, args)
+        with _lock:
+            evicted = _evict_locked()
+            _counter[0] += 1
+            key = _counter[0]
+            _local[key] = (self._tensor, time.monotonic() + _TTL, func, args)
+        for ent in evicted:
+            _release_producer_ref(ent[2], ent[3])
         return (_materialize, (os.getpid(), key, func, args))
 
     def tensor(self):
