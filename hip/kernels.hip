@@ -337,6 +337,7 @@ __global__ void conv1_u8_nhwc_kernel(const uint8_t* __restrict__ in,   // [N,4,H
 
 }  // namespace
 
+#include "conv3x3.hip.inc"
 #include "lstm.hip.inc"
 
 // ------------------------------------------------------------ wrappers
@@ -703,6 +704,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_add2_fwd", &bias_add2_fwd, "fused residual close: x + bias1 + shortcut (+ bias2)");
   m.def("frames_u8_to_bf16_nhwc", &frames_u8_to_bf16_nhwc, "fused uint8->bf16 NHWC scale");
   m.def("conv1_u8_nhwc", &conv1_u8_nhwc, "fused uint8 frames -> conv(4->16,3x3)+bias, NHWC bf16");
+  m.def("conv3x3_nhwc_fused", &conv3x3_nhwc_fused,
+        "MFMA 3x3/s1/p1 NHWC conv with fused relu/bias prologue + bias/relu/residual epilogue",
+        py::arg("x"), py::arg("w_packed"), py::arg("k"), py::arg("relu_in") = false,
+        py::arg("bias_in") = py::none(), py::arg("epi") = 0, py::arg("bias1") = py::none(),
+        py::arg("res") = py::none(), py::arg("bias2") = py::none());
   m.def("lstm_fused_fwd", &lstm_fused_fwd, "fused masked LSTM sequence scan fwd (MFMA, gfx950)");
   m.def("lstm_fused_bwd", &lstm_fused_bwd, "fused masked LSTM sequence scan bwd (MFMA, gfx950)");
   m.def("vtrace_from_log_rhos", &vtrace_from_log_rhos, "fused V-trace scan (gfx950)");

@@ -33,8 +33,31 @@ class ResidualBlock(nn.Module):
         (ops/fused_bias): convs run bias-free, relu/add are one pass each.
         `pending_bias` is a per-channel bias the CALLER still owes `x`
         (section conv bias carried through the maxpool).
+
+        Inference fast path (actor, no_grad): the whole block is TWO fused
+        MFMA conv kernels (ops/conv3x3) — producer relu+bias on load,
+        bias+residual on store.
         """
+        from moolib_amd.ops import conv3x3 as c3
         from moolib_amd.ops.fused_bias import bias_add2, bias_relu
+
+        ch = self.conv0.out_channels
+        if (
+            not torch.is_grad_enabled()
+            and x.is_cuda
+            and c3.available(x.size(1), ch)
+            and (x.dtype == torch.bfloat16 or torch.is_autocast_enabled())
+        ):
+            u = c3.conv3x3(
+                x, c3.pack_weight(self.conv0.weight), ch,
+                relu_in=True, bias_in=pending_bias,
+            )
+            return c3.conv3x3(
+                u, c3.pack_weight(self.conv1.weight), ch,
+                relu_in=True, bias_in=self.conv0.bias,
+                epi=c3.EPI_BIAS_ADD, bias1=self.conv1.bias,
+                res=x, bias2=pending_bias,
+            )
 
         if pending_bias is not None:
             t = bias_relu(x, pending_bias)
@@ -61,7 +84,18 @@ class ConvSection(nn.Module):
         if fused_bias.available(x, self.conv.out_channels):
             # bias-free conv; per-channel bias commutes with the per-channel
             # spatial max, so it rides into res0 as a pending bias.
-            u = F.conv2d(x, self.conv.weight, None, padding=1)
+            from moolib_amd.ops import conv3x3 as c3
+
+            if (
+                not torch.is_grad_enabled()
+                and c3.available(x.size(1), self.conv.out_channels)
+                and (x.dtype == torch.bfloat16 or torch.is_autocast_enabled())
+            ):
+                u = c3.conv3x3(
+                    x, c3.pack_weight(self.conv.weight), self.conv.out_channels
+                )
+            else:
+                u = F.conv2d(x, self.conv.weight, None, padding=1)
             u = maxpool3x3s2(u)
             u = self.res0.forward_fused(u, self.conv.bias)
             return self.res1.forward_fused(u)

@@ -1,0 +1,71 @@
+"""Fused MFMA 3x3 NHWC conv for the actor inference path.
+
+Wraps hip/conv3x3.hip.inc: implicit-GEMM conv on mfma_f32_16x16x32_bf16
+with the producer relu (+pending bias) fused into the input load and the
+bias / bias+relu / bias+residual epilogue fused into the store. An IMPALA
+residual block (reference examples/atari/models.py:30-42) becomes two
+kernel launches. Forward-only — gate on `not torch.is_grad_enabled()`.
+"""
+import os
+
+import torch
+
+from moolib_amd.ops.lstm import pack_mfma_b
+
+EPI_NONE, EPI_BIAS, EPI_BIAS_RELU, EPI_BIAS_ADD = 0, 1, 2, 3
+
+_kernels = None
+
+
+def _k():
+    global _kernels
+    if _kernels is None:
+        from moolib_amd import _kernels as k
+
+        _kernels = k
+    return _kernels
+
+
+def available(C, K):
+    if os.environ.get("MOOLIB_AMD_NO_CONV3_KERNEL"):
+        return False
+    if C not in (16, 32) or K not in (16, 32):
+        return False
+    try:
+        _k()
+    except ImportError:
+        return False
+    return True
+
+
+def pack_weight(w):
+    """nn.Conv2d weight [K, C, 3, 3] -> packed MFMA B fragments.
+
+    GEMM B is W[kd][kout] with kd = (kh*3+kw)*C + c, zero-padded so the
+    Kdim is a multiple of 32, then laid out in pack_mfma_b fragment order.
+    """
+    K, C, kh, kw = w.shape
+    assert kh == 3 and kw == 3
+    gem = w.permute(2, 3, 1, 0).reshape(9 * C, K)
+    kk32 = (9 * C + 31) // 32 * 32
+    if kk32 != 9 * C:
+        gem = torch.cat([gem, gem.new_zeros(kk32 - 9 * C, K)])
+    return pack_mfma_b(gem.to(torch.bfloat16).contiguous())
+
+
+def conv3x3(x, w_packed, k, relu_in=False, bias_in=None, epi=EPI_NONE,
+            bias1=None, res=None, bias2=None):
+    """out = conv3x3_s1_p1( relu(x + bias_in) if relu_in else x ) then
+    epilogue: none / +bias1 / relu(+bias1) / +bias1+res(+bias2)."""
+    to_bf = lambda t: None if t is None else t.to(torch.bfloat16)
+    return _k().conv3x3_nhwc_fused(
+        x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last),
+        w_packed,
+        k,
+        relu_in,
+        to_bf(bias_in),
+        epi,
+        to_bf(bias1),
+        None if res is None else to_bf(res).contiguous(memory_format=torch.channels_last),
+        to_bf(bias2),
+    )

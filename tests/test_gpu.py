@@ -445,3 +445,97 @@ class TestFusedBias:
             dg = (g_f[n] - g_e[n]).abs().max().item()
             rel = dg / (g_e[n].abs().max().item() + 1e-8)
             assert rel < 1e-2, (n, dg, rel)
+
+
+@gpu
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs MI355X")
+class TestConv3x3Fused:
+    """MFMA implicit-GEMM 3x3 conv vs fp32 F.conv2d (bf16 tolerance)."""
+
+    def _ref(self, x, w, relu_in=False, bias_in=None, epi=0, bias1=None, res=None, bias2=None):
+        import torch.nn.functional as F
+
+        xf = x.float()
+        if relu_in:
+            if bias_in is not None:
+                xf = xf + bias_in.float().view(1, -1, 1, 1)
+            xf = torch.relu(xf)
+        y = F.conv2d(xf, w.float(), None, padding=1)
+        if epi == 1:
+            y = y + bias1.float().view(1, -1, 1, 1)
+        elif epi == 2:
+            y = torch.relu(y + bias1.float().view(1, -1, 1, 1))
+        elif epi == 3:
+            y = y + bias1.float().view(1, -1, 1, 1) + res.float()
+            if bias2 is not None:
+                y = y + bias2.float().view(1, -1, 1, 1)
+        return y
+
+    @pytest.mark.parametrize("C,K", [(16, 16), (16, 32), (32, 32)])
+    def test_plain_conv(self, C, K):
+        from moolib_amd.ops import conv3x3 as c3
+
+        torch.manual_seed(C * 100 + K)
+        x = torch.randn(3, C, 11, 13, device="cuda", dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last
+        )
+        w = torch.randn(K, C, 3, 3, device="cuda") * 0.2
+        y = c3.conv3x3(x, c3.pack_weight(w), K)
+        ref = self._ref(x, w)
+        err = (y.float() - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err / scale < 0.02, (err, scale)
+
+    def test_fused_block_variants(self):
+        from moolib_amd.ops import conv3x3 as c3
+
+        torch.manual_seed(7)
+        C = K = 32
+        x = torch.randn(2, C, 21, 21, device="cuda", dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last
+        )
+        w = torch.randn(K, C, 3, 3, device="cuda") * 0.2
+        b_in = torch.randn(C, device="cuda") * 0.5
+        b1 = torch.randn(K, device="cuda") * 0.5
+        b2 = torch.randn(K, device="cuda") * 0.5
+        res = torch.randn(2, K, 21, 21, device="cuda", dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last
+        )
+        wp = c3.pack_weight(w)
+        for kwargs in (
+            dict(relu_in=True),
+            dict(relu_in=True, bias_in=b_in),
+            dict(epi=c3.EPI_BIAS, bias1=b1),
+            dict(epi=c3.EPI_BIAS_RELU, bias1=b1),
+            dict(relu_in=True, bias_in=b_in, epi=c3.EPI_BIAS_ADD, bias1=b1, res=res, bias2=b2),
+        ):
+            y = c3.conv3x3(x, wp, K, **kwargs)
+            ref = self._ref(x, w, **kwargs)
+            err = (y.float() - ref).abs().max().item()
+            scale = max(ref.abs().max().item(), 1.0)
+            assert err / scale < 0.02, (kwargs, err, scale)
+
+    def test_actor_model_path_matches(self):
+        """AtariNet no_grad forward: conv3x3 path vs MOOLIB_AMD_NO_CONV3_KERNEL."""
+        import os
+
+        from moolib_amd.models.atari import AtariNet
+
+        torch.manual_seed(11)
+        model = AtariNet(num_actions=6).to("cuda").to(torch.bfloat16)
+        model = model.to(memory_format=torch.channels_last)
+        inputs = {
+            "state": torch.randint(0, 255, (1, 16, 4, 84, 84), dtype=torch.uint8, device="cuda"),
+            "reward": torch.randn(1, 16, device="cuda"),
+            "prev_action": torch.randint(0, 6, (1, 16), device="cuda"),
+            "done": torch.zeros(1, 16, dtype=torch.bool, device="cuda"),
+        }
+        with torch.no_grad():
+            out_f, _ = model(inputs, tuple())
+            os.environ["MOOLIB_AMD_NO_CONV3_KERNEL"] = "1"
+            try:
+                out_e, _ = model(inputs, tuple())
+            finally:
+                del os.environ["MOOLIB_AMD_NO_CONV3_KERNEL"]
+        d = (out_f["policy_logits"].float() - out_e["policy_logits"].float()).abs().max().item()
+        assert d < 0.3, d  # bf16 accumulation-order differences through the net
