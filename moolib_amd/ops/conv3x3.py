@@ -27,7 +27,12 @@ def _k():
 
 
 def available(C, K):
-    if os.environ.get("MOOLIB_AMD_NO_CONV3_KERNEL"):
+    # Opt-in: measured on MI355X the v1 kernel (16-pixel M-tiles, no LDS
+    # tap reuse) is ~20% slower than MIOpen's tuned igemm on the actor path
+    # (44.4k vs 50.5k f/s same-box A/B) despite the fusion wins. Numerics
+    # are validated (tests/test_gpu.py::TestConv3x3Fused); revisit with
+    # LDS-staged input tiles + larger M-tiles before defaulting on.
+    if not os.environ.get("MOOLIB_AMD_CONV3_KERNEL"):
         return False
     if C not in (16, 32) or K not in (16, 32):
         return False

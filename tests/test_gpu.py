@@ -516,11 +516,12 @@ class TestConv3x3Fused:
             assert err / scale < 0.02, (kwargs, err, scale)
 
     def test_actor_model_path_matches(self):
-        """AtariNet no_grad forward: conv3x3 path vs MOOLIB_AMD_NO_CONV3_KERNEL."""
+        """AtariNet no_grad forward: conv3x3 path (opt-in env) vs MIOpen."""
         import os
 
         from moolib_amd.models.atari import AtariNet
 
+        os.environ["MOOLIB_AMD_CONV3_KERNEL"] = "1"
         torch.manual_seed(11)
         model = AtariNet(num_actions=6).to("cuda").to(torch.bfloat16)
         model = model.to(memory_format=torch.channels_last)
@@ -530,12 +531,12 @@ class TestConv3x3Fused:
             "prev_action": torch.randint(0, 6, (1, 16), device="cuda"),
             "done": torch.zeros(1, 16, dtype=torch.bool, device="cuda"),
         }
-        with torch.no_grad():
-            out_f, _ = model(inputs, tuple())
-            os.environ["MOOLIB_AMD_NO_CONV3_KERNEL"] = "1"
-            try:
+        try:
+            with torch.no_grad():
+                out_f, _ = model(inputs, tuple())
+                del os.environ["MOOLIB_AMD_CONV3_KERNEL"]
                 out_e, _ = model(inputs, tuple())
-            finally:
-                del os.environ["MOOLIB_AMD_NO_CONV3_KERNEL"]
+        finally:
+            os.environ.pop("MOOLIB_AMD_CONV3_KERNEL", None)
         d = (out_f["policy_logits"].float() - out_e["policy_logits"].float()).abs().max().item()
         assert d < 0.3, d  # bf16 accumulation-order differences through the net
