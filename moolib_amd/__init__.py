@@ -139,7 +139,10 @@ class _BatchCollector:
         self._last_arrival = None
 
     def _window(self):
-        if not self.dynamic or self.proc_ema is None:
+        # Queue-style handlers only enqueue (service time ~0, invisible to
+        # us): keep the full window so batches still form. Direct handlers
+        # adapt the window to their measured service time.
+        if not self.dynamic or self.proc_ema is None or self.proc_ema < 1e-3:
             return self.max_latency
         return min(max(self.proc_ema, self.min_latency), self.max_latency)
 
