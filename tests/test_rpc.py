@@ -325,3 +325,36 @@ class TestTransportModel:
         # the dialed tcp address has samples and a measured ema
         line = [l for l in info.splitlines() if "tm_host" in l and "transport=" in l][0]
         assert "ema=" in line and "n=5" in line, line
+
+
+class TestManyClientBatching:
+    """Many concurrent clients against one batched handler (reference
+    test/test_batch.py: batching define styles x N clients)."""
+
+    def test_40_clients_batched_sum(self):
+        host = moolib_amd.Rpc()
+        host.set_name("bhost")
+        addr = host.listen("127.0.0.1:0")[0]
+
+        def handler(x):
+            # batched: x is [B, 4]; reply is per-caller row sums [B]
+            assert x.dim() == 2 and x.size(1) == 4, x.shape
+            return x.sum(dim=1)
+
+        host.define("rowsum", handler, batch_size=8, dynamic_batching=True)
+
+        clients = []
+        for i in range(10):
+            c = moolib_amd.Rpc()
+            c.set_name("bc%d" % i)
+            c.set_timeout(30)
+            c.connect(addr)
+            clients.append(c)
+
+        futs = []
+        for rep in range(4):
+            for i, c in enumerate(clients):
+                x = torch.full((4,), float(i + rep * 10))
+                futs.append((i + rep * 10, c.async_("bhost", "rowsum", x)))
+        for want, f in futs:
+            assert float(f.result()) == want * 4.0
