@@ -358,3 +358,63 @@ class TestManyClientBatching:
                 futs.append((i + rep * 10, c.async_("bhost", "rowsum", x)))
         for want, f in futs:
             assert float(f.result()) == want * 4.0
+
+
+class TestSerdeBreadth:
+    """Wire-format coverage: every torch dtype the serializer maps, deep
+    nests, empty and non-contiguous tensors (csrc/serde.cc tag table)."""
+
+    def _pair(self):
+        host = moolib_amd.Rpc()
+        host.set_name("sd_host")
+        addr = host.listen("127.0.0.1:0")[0]
+        client = moolib_amd.Rpc()
+        client.set_name("sd_client")
+        client.set_timeout(20)
+        client.connect(addr)
+        host.define("echo", lambda x: x)
+        return host, client
+
+    def test_all_dtypes_roundtrip(self):
+        host, client = self._pair()
+        dtypes = [
+            torch.float32, torch.float64, torch.float16, torch.bfloat16,
+            torch.int64, torch.int32, torch.int16, torch.int8,
+            torch.uint8, torch.bool,
+        ]
+        for dt in dtypes:
+            if dt == torch.bool:
+                t = torch.tensor([True, False, True])
+            elif dt.is_floating_point:
+                t = torch.randn(3, 4).to(dt)
+            else:
+                t = torch.arange(12, dtype=dt).reshape(3, 4)
+            r = client.sync("sd_host", "echo", t)
+            assert r.dtype == dt and torch.equal(r.float(), t.float()), dt
+
+    def test_empty_and_noncontiguous(self):
+        host, client = self._pair()
+        e = torch.empty(0, 5)
+        r = client.sync("sd_host", "echo", e)
+        assert r.shape == (0, 5)
+        nc = torch.randn(6, 8).t()  # non-contiguous view
+        r = client.sync("sd_host", "echo", nc)
+        assert torch.equal(r, nc)
+
+    def test_deep_nest(self):
+        host, client = self._pair()
+        payload = {
+            "a": [1, 2.5, "s", None, True],
+            "b": {"c": (torch.ones(2), {"d": [torch.zeros(1), b"bytes"]})},
+            "e": [[[42]]],
+        }
+        r = client.sync("sd_host", "echo", payload)
+        assert r["a"] == [1, 2.5, "s", None, True]
+        assert torch.equal(r["b"]["c"][0], torch.ones(2))
+        assert r["b"]["c"][1]["d"][1] == b"bytes"
+        assert r["e"] == [[[42]]]
+
+    def test_unicode_names_and_args(self):
+        host, client = self._pair()
+        host.define("héllo🙂", lambda s: s + "!")
+        assert client.sync("sd_host", "héllo🙂", "ünïcodé") == "ünïcodé!"
