@@ -168,3 +168,27 @@ class TestForkHygiene:
         assert client.sync("fd_host", "hi") == 1  # live conns on both ends
         after = self._worker_socket_count()
         assert after <= base, (base, after)
+
+
+class FailsAtStep3Env:
+    def __init__(self):
+        self.n = 0
+
+    def reset(self):
+        return np.zeros(2, dtype=np.float32)
+
+    def step(self, action):
+        self.n += 1
+        if self.n == 3:
+            raise RuntimeError("env exploded at step 3")
+        return np.zeros(2, dtype=np.float32), 1.0, False, {}
+
+
+class TestMidEpisodeError:
+    def test_step_error_surfaces_not_hangs(self):
+        pool = moolib_amd.EnvPool(FailsAtStep3Env, num_processes=1, batch_size=2, num_batches=1)
+        import pytest as _pt
+
+        with _pt.raises(RuntimeError, match="exploded|died|failed"):
+            for i in range(10):
+                pool.step(0, torch.zeros(2, dtype=torch.int64)).result()
