@@ -74,7 +74,13 @@ void ReduceValue::fold(ReduceValue& src, const std::function<void(ReduceValue&, 
       numSkipped += src.numSkipped;
       if (tensors.size() != src.tensors.size())
         throw RpcError("allreduce: gradient bundle tensor count mismatch");
-      for (size_t i = 0; i < tensors.size(); ++i) tensors[i].add_(src.tensors[i]);
+      for (size_t i = 0; i < tensors.size(); ++i) {
+        // add_ would silently BROADCAST a mismatched shape (e.g. a peer on
+        // a different model revision) into corrupt gradients; refuse.
+        if (!tensors[i].sizes().equals(src.tensors[i].sizes()))
+          throw RpcError("allreduce: gradient tensor shape mismatch at index " + std::to_string(i));
+        tensors[i].add_(src.tensors[i]);
+      }
       break;
     case leaderTuple:
       if (std::tie(src.version, src.leaderName) > std::tie(version, leaderName)) {
