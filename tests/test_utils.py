@@ -53,7 +53,7 @@ class TestNest:
     def test_map_many(self):
         a = {"x": 1, "y": (2, 3)}
         b = {"x": 10, "y": (20, 30)}
-        s = nest.map_many(lambda u, v: u + v, a, b)
+        s = nest.map_many(lambda leaves: leaves[0] + leaves[1], a, b)
         assert s == {"x": 11, "y": (22, 33)}
 
     def test_preserves_types(self):
@@ -105,5 +105,5 @@ class TestBatchSizeFinder:
         def fn(batch):
             _t.sleep(0.001 + len(batch) * 2e-6)
 
-        best = find(fn, make_batch=lambda bs: list(range(bs)), max_batch_size=256)
-        assert best >= 32, best
+        best, results = find(fn, make_batch=lambda bs: list(range(bs)), max_batch_size=256)
+        assert best >= 32, (best, results)
