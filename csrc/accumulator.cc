@@ -248,7 +248,9 @@ void Accumulator::update() {
         buffers_[i].copy_(pendingBuffers_[i], /*non_blocking=*/true);
       }
       modelVersion_ = pendingVersion_;
-      hasNewState_ = true;
+      // A bare weight broadcast (empty payload, no state tensors) adopts
+      // silently; only a real state transfer surfaces through has_new_state.
+      hasNewState_ = !pendingStatePayload_.empty() || !pendingStateTensors_.empty();
       phase_ = Phase::running;
       MRL_LOG_INFO("accumulator '%s': model adopted at version %lld", name_.c_str(),
                    (long long)modelVersion_);
@@ -445,18 +447,43 @@ void Accumulator::applyPendingLocked() {
 }
 
 void Accumulator::maybeSendModelUpdatesLocked() {
-  if (!isLeader_ || buffers_.empty()) return;
-  if (secondsSince(lastBuffersBroadcast_) < 10.0) return;
-  lastBuffersBroadcast_ = now();
-  std::vector<at::Tensor> cpuBuffers;
-  {
-    NoGrad ng;
-    for (auto& b : buffers_) cpuBuffers.push_back(b.detach().to(at::kCPU));
+  if (!isLeader_) return;
+  if (!buffers_.empty() && secondsSince(lastBuffersBroadcast_) >= 10.0) {
+    lastBuffersBroadcast_ = now();
+    std::vector<at::Tensor> cpuBuffers;
+    {
+      NoGrad ng;
+      for (auto& b : buffers_) cpuBuffers.push_back(b.detach().to(at::kCPU));
+    }
+    for (auto& m : group_->members()) {
+      if (m == group_->myName()) continue;
+      rpc_->sendRequest(m, fn("buffers"), "", cpuBuffers,
+                        [](Frame*, const std::string*) {}, 10.0);
+    }
   }
-  for (auto& m : group_->members()) {
-    if (m == group_->myName()) continue;
-    rpc_->sendRequest(m, fn("buffers"), "", cpuBuffers,
-                      [](Frame*, const std::string*) {}, 10.0);
+  // Periodic full-model broadcast (drift correction; the reference sends
+  // full state every 600 s, accumulator.cc:744-768). Params+buffers only —
+  // user state (optimizer etc.) still flows through the requestModel path.
+  // An empty state payload means "adopt weights silently" on the receiver.
+  if (lastModelBroadcast_.time_since_epoch().count() == 0) lastModelBroadcast_ = now();
+  if (secondsSince(lastModelBroadcast_) >= 600.0) {
+    lastModelBroadcast_ = now();
+    std::vector<at::Tensor> frameTensors;
+    {
+      NoGrad ng;
+      for (auto& p : allParams_) frameTensors.push_back(p.detach().to(at::kCPU));
+      for (auto& b : buffers_) frameTensors.push_back(b.detach().to(at::kCPU));
+    }
+    WireWriter w;
+    w.i64(modelVersion_);
+    w.u32(static_cast<uint32_t>(allParams_.size()));
+    w.u32(static_cast<uint32_t>(buffers_.size()));
+    w.str("");
+    for (auto& m : group_->members()) {
+      if (m == group_->myName()) continue;
+      rpc_->sendRequest(m, fn("modelupd"), w.out, frameTensors,
+                        [](Frame*, const std::string*) {}, 30.0);
+    }
   }
 }
 

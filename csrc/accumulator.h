@@ -132,6 +132,7 @@ class Accumulator : public std::enable_shared_from_this<Accumulator> {
   // leader side
   std::vector<std::string> stateRequesters_;
   TimePoint lastBuffersBroadcast_{};
+  TimePoint lastModelBroadcast_{};
 
   // gradient machine: parallelGradients_ staging slots, used round-robin.
   // Count rounds are strictly sequential cluster-wide (decided via shared
