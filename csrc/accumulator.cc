@@ -3,6 +3,7 @@
 #include <ATen/Functions.h>
 #include <c10/core/GradMode.h>
 
+#include <cstdlib>
 #include <sstream>
 
 namespace mrl {
@@ -42,6 +43,9 @@ Accumulator::Accumulator(std::string name, std::vector<at::Tensor> params,
   }
   slots_.resize(1);
   slots_[0].flat = makeFlatLocked();
+  if (const char* e = getenv("MOOLIB_AMD_MODEL_BCAST_S")) {  // test hook
+    modelBcastInterval_ = atof(e);
+  }
 }
 
 at::Tensor Accumulator::makeFlatLocked() {
@@ -466,7 +470,7 @@ void Accumulator::maybeSendModelUpdatesLocked() {
   // user state (optimizer etc.) still flows through the requestModel path.
   // An empty state payload means "adopt weights silently" on the receiver.
   if (lastModelBroadcast_.time_since_epoch().count() == 0) lastModelBroadcast_ = now();
-  if (secondsSince(lastModelBroadcast_) >= 600.0) {
+  if (secondsSince(lastModelBroadcast_) >= modelBcastInterval_) {
     lastModelBroadcast_ = now();
     std::vector<at::Tensor> frameTensors;
     {

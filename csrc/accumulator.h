@@ -133,6 +133,7 @@ class Accumulator : public std::enable_shared_from_this<Accumulator> {
   std::vector<std::string> stateRequesters_;
   TimePoint lastBuffersBroadcast_{};
   TimePoint lastModelBroadcast_{};
+  double modelBcastInterval_ = 600.0;  // MOOLIB_AMD_MODEL_BCAST_S overrides
 
   // gradient machine: parallelGradients_ staging slots, used round-robin.
   // Count rounds are strictly sequential cluster-wide (decided via shared

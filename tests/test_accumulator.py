@@ -340,3 +340,31 @@ class TestParallelGradients:
             # the multiset of applied averages must match exactly
             assert sorted(applied[i][:rounds]) == want, applied[i]
         assert overlapped[0], "no pipelining observed"
+
+
+class TestPeriodicModelBroadcast:
+    def test_leader_rebroadcast_adopts_silently(self, monkeypatch):
+        """The leader periodically re-broadcasts weights (drift correction);
+        followers adopt them WITHOUT surfacing has_new_state (that path is
+        for real user-state transfers)."""
+        monkeypatch.setenv("MOOLIB_AMD_MODEL_BCAST_S", "0.5")
+        c = AccCluster(2)
+        c.wait_connected()
+        leader_name = c.peers[0].acc.get_leader()
+        leader = next(p for p in c.peers if p.rpc.get_name() == leader_name)
+        follower = next(p for p in c.peers if p.rpc.get_name() != leader_name)
+        # any join-time state transfer settles first
+        c.pump(lambda: follower.state_got is not None or not follower.acc.has_new_state(),
+               deadline=5)
+        follower.state_got = None
+        # desync the follower's weights; the periodic broadcast must repair
+        with torch.no_grad():
+            follower.params[0].add_(1000.0)
+        want = leader.params[0].detach().clone()
+        ok = c.pump(
+            lambda: torch.allclose(follower.params[0].detach(), want, atol=1e-5),
+            deadline=15,
+        )
+        assert ok, "weights were not repaired by the periodic broadcast"
+        # silent adoption: no new user state surfaced
+        assert follower.state_got is None
