@@ -420,6 +420,14 @@ class ImpalaPeer:
         )
         self.optimizer.step()
         self._sync_shadow()
+        # Refresh cached MFMA-packed conv weights (ops/conv3x3) so actor
+        # graph replays read the updated weights; capturable ops only.
+        from moolib_amd.ops import conv3x3 as _c3
+
+        import os as _os
+
+        if _os.environ.get("MOOLIB_AMD_CONV3_KERNEL"):
+            _c3.repack(self.fwd_model)
         return norm
 
     def step_optimizer(self):

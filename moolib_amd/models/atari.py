@@ -49,11 +49,11 @@ class ResidualBlock(nn.Module):
             and (x.dtype == torch.bfloat16 or torch.is_autocast_enabled())
         ):
             u = c3.conv3x3(
-                x, c3.pack_weight(self.conv0.weight), ch,
+                x, c3.packed_buffer(self.conv0), ch,
                 relu_in=True, bias_in=pending_bias,
             )
             return c3.conv3x3(
-                u, c3.pack_weight(self.conv1.weight), ch,
+                u, c3.packed_buffer(self.conv1), ch,
                 relu_in=True, bias_in=self.conv0.bias,
                 epi=c3.EPI_BIAS_ADD, bias1=self.conv1.bias,
                 res=x, bias2=pending_bias,
@@ -92,7 +92,7 @@ class ConvSection(nn.Module):
                 and (x.dtype == torch.bfloat16 or torch.is_autocast_enabled())
             ):
                 u = c3.conv3x3(
-                    x, c3.pack_weight(self.conv.weight), self.conv.out_channels
+                    x, c3.packed_buffer(self.conv), self.conv.out_channels
                 )
             else:
                 u = F.conv2d(x, self.conv.weight, None, padding=1)

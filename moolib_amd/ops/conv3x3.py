@@ -58,6 +58,41 @@ def pack_weight(w):
     return pack_mfma_b(gem.to(torch.bfloat16).contiguous())
 
 
+def packed_buffer(conv):
+    """Version-checked persistent packed-weight buffer for an nn.Conv2d.
+
+    Packing costs ~6 small kernels per conv; done inline it dominates the
+    graphed actor forward (measured: the kernel itself BEATS MIOpen at
+    actor shapes in isolation — profiles/r01_conv3x3_micro.txt — but
+    per-replay repacking erased the win). The buffer has a stable address,
+    so hipGraphs may capture reads of it; refresh after weight updates
+    with repack() (graph-capturable: pack + copy_ into the buffer).
+    """
+    w = conv.weight
+    ver = w._version
+    cache = getattr(conv, "_c3_cache", None)
+    if cache is None:
+        conv._c3_cache = cache = [ver, pack_weight(w.detach())]
+        return cache[1]
+    if cache[0] != ver:
+        cache[1].copy_(pack_weight(w.detach()))
+        cache[0] = ver
+    return cache[1]
+
+
+def repack(module):
+    """Refresh every cached packed weight in `module` (call after the
+    optimizer step; safe inside a hipGraph capture)."""
+    import torch.nn as nn
+
+    for m in module.modules():
+        if isinstance(m, nn.Conv2d):
+            cache = getattr(m, "_c3_cache", None)
+            if cache is not None:
+                cache[1].copy_(pack_weight(m.weight.detach()))
+                cache[0] = m.weight._version
+
+
 def conv3x3(x, w_packed, k, relu_in=False, bias_in=None, epi=EPI_NONE,
             bias1=None, res=None, bias2=None):
     """out = conv3x3_s1_p1( relu(x + bias_in) if relu_in else x ) then
