@@ -34,3 +34,19 @@ def test_bench_two_rank_gloo(tmp_path):
     assert j["n_gpus"] == 2 and j["steps"] == 3
     assert j["scaling"] == "weak" and j["value"] > 0
     assert j["config"]["parallelism"] == "dp2"
+
+
+@pytest.mark.timeout(180)
+def test_bench_r2d2_config_json(tmp_path):
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "bench.py"), "--config", "r2d2", "--steps", "8"],
+        cwd=str(tmp_path), capture_output=True, text=True, timeout=170,
+        env={**os.environ, "PYTHONPATH": root},
+    )
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    j = json.loads(lines[0])
+    assert "replay" in j["metric"] and j["unit"] == "sequences/s"
+    assert j["config"]["capacity_seqs"] == 2048
