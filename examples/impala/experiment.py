@@ -42,12 +42,27 @@ def parse_args():
     return ap.parse_args()
 
 
+def _yaml_value(v):
+    out = yaml.safe_load(v)
+    # PyYAML (YAML 1.1) parses "2e9"/"6e-4" as STRINGS (needs "2.0e+9");
+    # users write scientific notation constantly, so coerce numeric-looking
+    # strings here.
+    if isinstance(out, str):
+        for cast in (int, float):
+            try:
+                return cast(out)
+            except ValueError:
+                pass
+    return out
+
+
 def load_flags(args):
     with open(args.config) as f:
         flags = yaml.safe_load(f)
+    flags = {k: (_yaml_value(v) if isinstance(v, str) else v) for k, v in flags.items()}
     for kv in args.override:
         k, v = kv.split("=", 1)
-        flags[k] = yaml.safe_load(v)
+        flags[k] = _yaml_value(v)
     for k in ("connect", "device", "savedir", "total_steps"):
         v = getattr(args, k.replace("-", "_"), None)
         if v is not None:

@@ -33,6 +33,18 @@ from moolib_amd.utils.stats import GlobalStatsAccumulator, StatMean, StatSum
 
 @dataclasses.dataclass
 class ImpalaConfig:
+    def __post_init__(self):
+        # Coerce numeric fields that arrived as strings (e.g. yaml "6e-4",
+        # which YAML 1.1 parses as a string) to the declared field type.
+        for f in dataclasses.fields(self):
+            v = getattr(self, f.name)
+            if isinstance(v, str) and f.type in ("int", "float", int, float):
+                cast = int if f.type in ("int", int) else float
+                try:
+                    setattr(self, f.name, cast(float(v)))
+                except ValueError:
+                    pass
+
     num_actions: int = 18
     actor_batch_size: int = 128
     num_actor_batches: int = 2
