@@ -68,7 +68,8 @@ class GraphedCall:
                 logging.warning("hipGraph capture failed for '%s' (%s); staying eager", self.name, e)
                 self.failed = True
                 self.graph = None
-                torch.cuda.synchronize()
+                if torch.cuda.is_available():
+                    torch.cuda.synchronize()  # leave no half-captured state behind
                 return self.fn(inputs)
         _copy_nest(self.static_in, inputs)
         self.graph.replay()
