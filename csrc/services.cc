@@ -297,6 +297,16 @@ void Group::update() {
                 g->active_ = false;
                 MRL_LOG_INFO("group '%s': lost contact with broker (%s)", g->name_.c_str(),
                              error->c_str());
+              } else if (!g->active_ && g->lastPingOk_.time_since_epoch().count() == 0 &&
+                         secondsSince(g->lastUnreachableWarn_) > 10.0) {
+                // Never reached the broker at all: almost always a wrong
+                // broker NAME (set_broker_name; the peer must be literally
+                // named that) or address. Silent forever is undebuggable.
+                g->lastUnreachableWarn_ = now();
+                MRL_LOG_ERROR(
+                    "group '%s': cannot reach broker peer '%s' (%s) — check the broker's "
+                    "rpc name and connect() address",
+                    g->name_.c_str(), g->brokerName_.c_str(), error->c_str());
               }
               return;
             }

@@ -165,6 +165,7 @@ class Group : public std::enable_shared_from_this<Group> {
   bool pingInFlight_ = false;
   TimePoint lastPing_{};
   TimePoint lastPingOk_{};
+  TimePoint lastUnreachableWarn_{};
   std::unordered_map<std::string, uint64_t> seqByName_;  // per-opName sequence
   std::unordered_map<OpKey, AllReduceOpPtr, OpKeyHash> ops_;
   bool stopped_ = false;
