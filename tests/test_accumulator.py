@@ -368,3 +368,32 @@ class TestPeriodicModelBroadcast:
         assert ok, "weights were not repaired by the periodic broadcast"
         # silent adoption: no new user state surfaced
         assert follower.state_got is None
+
+
+class TestSoloPeer:
+    def test_single_member_full_cycle(self):
+        """One peer + broker: election, counting, reduction, and gradient
+        application must all work with a member list of one (a2c's shape)."""
+        c = AccCluster(1, virtual_batch_size=4)
+        c.wait_connected()
+        p = c.peers[0]
+        steps = 0
+        t0 = time.time()
+        while steps < 3 and time.time() - t0 < 30:
+            c.broker.update()
+            p.pump_once()
+            acc = p.acc
+            if not acc.connected():
+                time.sleep(0.005)
+                continue
+            if acc.has_gradients():
+                assert p.params[0].grad is not None
+                steps += 1
+                acc.zero_gradients()
+            elif acc.wants_gradients():
+                (p.params[0] * 2).sum().backward()
+                acc.reduce_gradients(4)
+            else:
+                time.sleep(0.001)
+        assert steps == 3, p.acc.debug_state()
+        assert acc.is_leader()
