@@ -436,6 +436,49 @@ void Rpc::sendRequest(const std::string& peerName, const std::string& funcName, 
   rec.frame = f;
   rec.deadline = now() + std::chrono::duration_cast<Clock::duration>(std::chrono::duration<double>(timeoutS));
   rec.cb = std::move(cb);
+  if (peerName == name_) {
+    // Self-call: a peer addressing itself by name dispatches locally — no
+    // wire, no connection (the reference instead detects and closes
+    // self-connections; local dispatch keeps "peer lists that include
+    // yourself" uniform). The Outgoing record stays registered so the
+    // normal timeout machinery still covers deferred handlers that never
+    // respond; completion flows through handleResponse with a synthetic
+    // frame (sentOn=0 -> no transport attribution).
+    Handler handler;
+    {
+      auto fit = functions_.find(f.fid);
+      if (fit != functions_.end()) handler = fit->second.second;
+    }
+    std::weak_ptr<Rpc> weak = weak_from_this();
+    RespondFn respond = [weak, rid](std::string payload, std::vector<at::Tensor> tensors,
+                                    bool isError) {
+      auto self = weak.lock();
+      if (!self || self->stopping_.load()) return;
+      Frame resp;
+      resp.kind = isError ? FrameKind::errorResponse : FrameKind::response;
+      resp.rid = rid;
+      resp.payload = std::move(payload);
+      resp.tensors = std::move(tensors);
+      self->handleResponse(0, std::move(resp), isError);
+    };
+    if (!handler) {
+      uint64_t fid = f.fid;
+      globalScheduler().run([respond, fid] {
+        respond("unknown function id " + std::to_string(fid), {}, true);
+      });
+    } else {
+      globalScheduler().run(
+          [handler = std::move(handler), f = std::move(f), selfName = name_,
+           respond = std::move(respond)]() mutable {
+            try {
+              handler(std::move(f), selfName, respond);
+            } catch (const std::exception& e) {
+              respond(std::string("handler exception: ") + e.what(), {}, true);
+            }
+          });
+    }
+    return;
+  }
   PeerInfo& p = getPeer(peerName);
   uint64_t fbytes = f.payload.size();
   for (auto& t : f.tensors) fbytes += t.nbytes();
@@ -640,7 +683,7 @@ void Rpc::timerLoop() {
           }
         } else {
           tryConnectPeerLocked(rec.peerName, p);
-          if (p.connecting == 0) broadcastFindPeerLocked(rec.peerName);
+          if (p.connecting == 0 && rec.peerName != name_) broadcastFindPeerLocked(rec.peerName);
         }
       }
       // Incoming GC.

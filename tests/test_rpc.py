@@ -418,3 +418,36 @@ class TestSerdeBreadth:
         host, client = self._pair()
         host.define("héllo🙂", lambda s: s + "!")
         assert client.sync("sd_host", "héllo🙂", "ünïcodé") == "ünïcodé!"
+
+
+class TestSelfCall:
+    """A peer calling itself by name dispatches locally (no connection)."""
+
+    def test_self_sync_and_async(self):
+        r = moolib_amd.Rpc()
+        r.set_name("me")
+        r.set_timeout(10)
+        r.listen("127.0.0.1:0")
+        r.define("double", lambda x: x * 2)
+        assert r.sync("me", "double", 21) == 42
+        f = r.async_("me", "double", torch.ones(3))
+        assert torch.equal(f.result(), torch.full((3,), 2.0))
+
+    def test_self_unknown_function(self):
+        r = moolib_amd.Rpc()
+        r.set_name("me2")
+        r.set_timeout(10)
+        with pytest.raises(moolib_amd.RpcError, match="does not exist"):
+            r.sync("me2", "missing")
+
+    def test_self_handler_exception(self):
+        r = moolib_amd.Rpc()
+        r.set_name("me3")
+        r.set_timeout(10)
+
+        def boom():
+            raise ValueError("kablam")
+
+        r.define("boom", boom)
+        with pytest.raises(moolib_amd.RpcError, match="kablam"):
+            r.sync("me3", "boom")
