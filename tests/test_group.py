@@ -149,3 +149,28 @@ class TestGroup:
         fut = g.all_reduce("x", 1)
         fut.wait(10)
         assert isinstance(fut.exception(), moolib_amd.RpcError)
+
+
+class TestCombinedBrokerMember:
+    def test_one_rpc_is_broker_and_member(self):
+        """Self-call dispatch lets a single Rpc host the Broker AND join a
+        group (pings to 'broker' == itself resolve locally)."""
+        rpc = moolib_amd.Rpc()
+        rpc.set_name("broker")
+        broker = moolib_amd.Broker(rpc)
+        rpc.listen("127.0.0.1:0")
+        g = moolib_amd.Group(rpc, "combined")
+        g.set_timeout(5)
+        t0 = time.time()
+        while time.time() - t0 < 15 and not g.active():
+            broker.update()
+            g.update()
+            time.sleep(0.02)
+        assert g.active() and g.members() == ["broker"]
+        fut = g.all_reduce("x", 7)
+        t0 = time.time()
+        while time.time() - t0 < 10 and not fut.done():
+            broker.update()
+            g.update()
+            time.sleep(0.02)
+        assert fut.result() == 7
