@@ -174,3 +174,13 @@ class TestCombinedBrokerMember:
             g.update()
             time.sleep(0.02)
         assert fut.result() == 7
+
+
+class TestWideGroup:
+    def test_twelve_peer_tree_allreduce(self):
+        """Deeper binary tree (12 members, 4 levels) with churn-free sums."""
+        c = Cluster(12)
+        c.wait_active()
+        for r in range(3):
+            res = c.reduce_all("wide", [r * 100 + i for i in range(12)])
+            assert res == [r * 1200 + 66] * 12
