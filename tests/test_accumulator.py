@@ -397,3 +397,29 @@ class TestSoloPeer:
                 time.sleep(0.001)
         assert steps == 3, p.acc.debug_state()
         assert acc.is_leader()
+
+
+class TestEightPeers:
+    def test_eight_peer_gradient_rounds(self):
+        """Control-plane at the 8-GPU scale shape: 8 members, virtual batch
+        8 (one contribution each), several reduction rounds."""
+        c = AccCluster(8, virtual_batch_size=8)
+        c.wait_connected()
+        steps = [0] * 8
+        t0 = time.time()
+        while min(steps) < 2 and time.time() - t0 < 60:
+            c.broker.update()
+            for i, p in enumerate(c.peers):
+                p.pump_once()
+                acc = p.acc
+                if not acc.connected():
+                    continue
+                if acc.has_gradients():
+                    gs = acc.get_gradient_stats()
+                    assert gs["batch_size"] >= 8 and gs["num_gradients"] >= 8
+                    steps[i] += 1
+                    acc.zero_gradients()
+                elif acc.wants_gradients():
+                    (p.params[0].sum() + p.params[1].sum()).backward()
+                    acc.reduce_gradients(1)
+        assert min(steps) >= 2, (steps, c.peers[0].acc.debug_state())
