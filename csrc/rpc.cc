@@ -280,6 +280,16 @@ void Rpc::handleGreeting(ConnId id, Frame& f) {
   }
 }
 
+void Rpc::handleResponseAck(ConnId id, Frame& f) {
+  std::lock_guard<std::mutex> lk(mu_);
+  auto cit = conns_.find(id);
+  if (cit == conns_.end() || cit->second.peerUid.empty()) return;
+  auto it = incoming_.find(IncomingKey{cit->second.peerUid, f.rid});
+  if (it == incoming_.end() || !it->second.responded) return;
+  it->second.acked = true;
+  it->second.response = Frame();  // free payload + tensors; marker stays for dedupe
+}
+
 // ----------------------------------------------------------- peer mgmt
 
 Rpc::PeerInfo& Rpc::getPeer(const std::string& name) { return peers_[name]; }
@@ -507,10 +517,10 @@ void Rpc::handleRequest(ConnId id, Frame&& f) {
     auto [iit, inserted] = incoming_.try_emplace(key);
     iit->second.lastConn = id;
     if (!inserted) {
-      if (iit->second.responded) {
+      if (iit->second.responded && !iit->second.acked) {
         engine_->send(id, iit->second.response);  // duplicate of a completed request
       }
-      return;  // duplicate of an in-flight request: drop
+      return;  // duplicate of an in-flight (or acked) request: drop
     }
     auto fit = functions_.find(f.fid);
     if (fit != functions_.end()) {
@@ -580,6 +590,16 @@ void Rpc::handleResponse(ConnId id, Frame&& f, bool isError) {
     for (auto& t : f.tensors) fbytes += t.nbytes();
     p.bytesRecv += fbytes;
     outgoing_.erase(it);
+    if (id != 0) {
+      // Ack the response so the responder can free its stored copy (it
+      // keeps full response frames — tensors included — for duplicate
+      // requests until acked or 60 s; without acks a busy tensor-serving
+      // peer retains every reply for the full minute).
+      Frame ackf;
+      ackf.kind = FrameKind::responseAck;
+      ackf.rid = f.rid;
+      engine_->send(id, std::move(ackf));
+    }
   }
   if (!cb) return;
   globalScheduler().run([cb = std::move(cb), f = std::move(f), isError, peerName, funcName]() mutable {
@@ -621,6 +641,9 @@ void Rpc::onFrame(ConnId id, Frame&& f) {
       break;
     case FrameKind::peerInfo:
       handlePeerInfo(id, f);
+      break;
+    case FrameKind::responseAck:
+      handleResponseAck(id, f);
       break;
     case FrameKind::keepalive:
       break;

@@ -136,7 +136,8 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
   };
   struct Incoming {
     bool responded = false;
-    Frame response;  // kept for re-send to duplicate requests
+    bool acked = false;  // sender confirmed receipt: stored response freed
+    Frame response;      // kept for re-send to duplicate requests until acked
     ConnId lastConn = 0;
     TimePoint doneTime{};
   };
@@ -159,6 +160,7 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
   void handleRequest(ConnId id, Frame&& f);
   void handleResponse(ConnId id, Frame&& f, bool isError);
   void handleFindPeer(ConnId id, Frame& f);
+  void handleResponseAck(ConnId id, Frame& f);
   void handlePeerInfo(ConnId id, Frame& f);
 
   void sendGreeting(ConnId id);
