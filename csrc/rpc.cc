@@ -281,13 +281,24 @@ void Rpc::handleGreeting(ConnId id, Frame& f) {
 }
 
 void Rpc::handleResponseAck(ConnId id, Frame& f) {
-  std::lock_guard<std::mutex> lk(mu_);
-  auto cit = conns_.find(id);
-  if (cit == conns_.end() || cit->second.peerUid.empty()) return;
-  auto it = incoming_.find(IncomingKey{cit->second.peerUid, f.rid});
-  if (it == incoming_.end() || !it->second.responded) return;
-  it->second.acked = true;
-  it->second.response = Frame();  // free payload + tensors; marker stays for dedupe
+  // Destroying py-backed tensors can ACQUIRE THE GIL (torch pyobj decref);
+  // doing that while holding mu_ on the reactor thread deadlocks against a
+  // python thread that holds the GIL and wants mu_. Move the frame out
+  // under the lock; let a scheduler thread (lock-free) destroy it.
+  Frame tomb;
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto cit = conns_.find(id);
+    if (cit == conns_.end() || cit->second.peerUid.empty()) return;
+    auto it = incoming_.find(IncomingKey{cit->second.peerUid, f.rid});
+    if (it == incoming_.end() || !it->second.responded) return;
+    it->second.acked = true;
+    tomb = std::move(it->second.response);
+    it->second.response = Frame();  // marker stays for dedupe
+  }
+  if (!tomb.payload.empty() || !tomb.tensors.empty()) {
+    globalScheduler().run([tomb = std::move(tomb)]() mutable {});
+  }
 }
 
 // ----------------------------------------------------------- peer mgmt
@@ -589,7 +600,11 @@ void Rpc::handleResponse(ConnId id, Frame&& f, bool isError) {
     uint64_t fbytes = f.payload.size();
     for (auto& t : f.tensors) fbytes += t.nbytes();
     p.bytesRecv += fbytes;
+    Frame reqTomb = std::move(it->second.frame);  // destroyed after unlock
     outgoing_.erase(it);
+    if (!reqTomb.tensors.empty()) {
+      globalScheduler().run([reqTomb = std::move(reqTomb)]() mutable {});
+    }
     if (id != 0) {
       // Ack the response so the responder can free its stored copy (it
       // keeps full response frames — tensors included — for duplicate
@@ -656,12 +671,17 @@ void Rpc::onFrame(ConnId id, Frame&& f) {
 
 void Rpc::failOutgoing(uint64_t rid, const std::string& error) {
   ResponseCallback cb;
+  Frame reqTomb;
   {
     std::lock_guard<std::mutex> lk(mu_);
     auto it = outgoing_.find(rid);
     if (it == outgoing_.end()) return;
     cb = std::move(it->second.cb);
+    reqTomb = std::move(it->second.frame);  // destroy outside mu_ (GIL hazard)
     outgoing_.erase(it);
+  }
+  if (!reqTomb.tensors.empty()) {
+    globalScheduler().run([reqTomb = std::move(reqTomb)]() mutable {});
   }
   if (!cb) return;
   globalScheduler().run([cb = std::move(cb), error] { cb(nullptr, &error); });
@@ -676,6 +696,7 @@ void Rpc::timerLoop() {
     if (stopping_.load()) return;
     TimePoint t = now();
     std::vector<std::pair<uint64_t, std::string>> failures;
+    std::vector<Frame> tombs;  // frames destroyed after mu_ is released
     {
       std::lock_guard<std::mutex> lk(mu_);
       // Reconnect persistent endpoints.
@@ -709,9 +730,12 @@ void Rpc::timerLoop() {
           if (p.connecting == 0 && rec.peerName != name_) broadcastFindPeerLocked(rec.peerName);
         }
       }
-      // Incoming GC.
+      // Incoming GC. Stored response frames may hold the last reference to
+      // py-backed tensors; their destruction can take the GIL, so it must
+      // happen outside mu_ (tombs vector, cleared after the lock scope).
       for (auto it = incoming_.begin(); it != incoming_.end();) {
         if (it->second.responded && secondsSince(it->second.doneTime) > 60.0) {
+          tombs.push_back(std::move(it->second.response));
           it = incoming_.erase(it);
         } else {
           ++it;
@@ -733,6 +757,7 @@ void Rpc::timerLoop() {
         ++it;
       }
     }
+    tombs.clear();  // may take the GIL (py-backed tensor decref) — no locks held
     for (auto& [rid, err] : failures) failOutgoing(rid, err);
   }
 }

@@ -69,6 +69,10 @@ void Scheduler::workerLoop() {
     } catch (...) {
       MRL_LOG_ERROR("scheduler task threw unknown exception");
     }
+    // Destroy the task's captures BEFORE re-locking: they may hold the last
+    // reference to python objects / py-backed tensors, whose destruction
+    // acquires the GIL — never block on the GIL while holding mu_.
+    f = nullptr;
     lk.lock();
     --busyThreads_;
     if (queue_.empty() && busyThreads_ == 0) idleCv_.notify_all();
