@@ -202,6 +202,10 @@ void Accumulator::update() {
   bool active = group_->active();
   uint64_t sid = group_->syncId();
 
+  // Hook poll closures capture python objects; their destruction can take
+  // the GIL, so collect them here and let them die AFTER mu_ is released
+  // (declared before lk: destroyed after it).
+  std::vector<std::function<bool()>> pollTombs;
   std::unique_lock<std::mutex> lk(mu_);
   // Drain abandoned hook collectives regardless of phase.
   for (auto it = drains_.begin(); it != drains_.end();) {
@@ -211,7 +215,12 @@ void Accumulator::update() {
     } catch (const std::exception& e) {
       MRL_LOG_ERROR("abandoned hook drain failed: %s", e.what());
     }
-    it = done ? drains_.erase(it) : ++it;
+    if (done) {
+      pollTombs.push_back(std::move(*it));
+      it = drains_.erase(it);
+    } else {
+      ++it;
+    }
   }
 
   if (!active) {
@@ -288,12 +297,14 @@ void Accumulator::update() {
         try {
           done = s.hookPoll();
         } catch (const std::exception& e) {
+          pollTombs.push_back(std::move(s.hookPoll));
           s.hookPoll = nullptr;
           MRL_LOG_ERROR("local reduce hook poll failed: %s", e.what());
           failAndResyncLocked("hook poll failed");
           return;
         }
         if (done) {
+          pollTombs.push_back(std::move(s.hookPoll));
           s.hookPoll = nullptr;
           NoGrad ng;
           at::Tensor result = s.flat.clone();
