@@ -650,8 +650,13 @@ void Accumulator::setParallelGradients(int64_t n) {
 }
 
 void Accumulator::setLocalReduceHook(LocalReduceHook h) {
-  std::lock_guard<std::mutex> lk(mu_);
-  hook_ = std::move(h);
+  LocalReduceHook old;
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    old = std::move(hook_);
+    hook_ = std::move(h);
+  }
+  // old (py-capturing) destroyed here, outside mu_
 }
 
 std::string Accumulator::debugState() {
