@@ -62,3 +62,35 @@ class TestRpcThroughput:
         gbps = n * x.nbytes / dt / 1e9
         print("64MiB tensor rpc: %.2f GB/s one-way payload" % gbps)
         assert gbps > 0.2, gbps  # loose floor
+
+
+def _rss_mb():
+    return int(open("/proc/self/status").read().split("VmRSS:")[1].split()[0]) / 1024
+
+
+class TestNoResponseRetention:
+    def test_tensor_replies_not_retained(self):
+        """Responders must free stored reply frames once the caller acks
+        (regression: replies were retained 60 s for dedupe -> a tensor
+        server pinned every reply for a minute)."""
+        import gc
+
+        host, client = make_pair()
+        host.define("echo", lambda t: t)
+        x = torch.randn(64, 64)  # 16 KiB payload
+        client.sync("host", "echo", x)
+        gc.collect()
+        r0 = _rss_mb()
+        n = 0
+        t0 = time.time()
+        while time.time() - t0 < 8:
+            futs = [client.async_("host", "echo", x) for _ in range(64)]
+            for f in futs:
+                f.result()
+            n += 64
+        gc.collect()
+        r1 = _rss_mb()
+        # without acks this grew ~16 KiB/call (~160 MB here); with them the
+        # 60 s rid markers cost ~100 B/call
+        limit = 30 + n * 0.001  # MB; generous for allocator noise
+        assert r1 - r0 < limit, (n, r0, r1)
