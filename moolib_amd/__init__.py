@@ -161,12 +161,15 @@ class _BatchCollector:
             self.pending.append((deferred, args, kwargs))
             n = len(self.pending)
             full = n >= self.batch_size
-            # Early flush: if filling the rest of the batch is expected to
-            # take longer than the window, waiting only adds latency.
+            # Early flush on the MARGINAL rule: if even the NEXT arrival is
+            # expected to take longer than the window, waiting only adds
+            # latency. (Comparing the time to FILL the batch would flush
+            # singles forever whenever batch_size exceeds the number of
+            # concurrent callers.)
             stale = (
                 self.dynamic
                 and self.arrival_ema is not None
-                and (self.batch_size - n) * self.arrival_ema > self._window()
+                and self.arrival_ema > self._window()
             )
             if full or stale:
                 fire = self.pending
