@@ -19,15 +19,15 @@ import moolib_amd
 from moolib_amd.models.atari import AtariNet
 
 
-def main():
-    ap = argparse.ArgumentParser()
-    ap.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
-    ap.add_argument("--clients", type=int, default=8)
-    ap.add_argument("--batch-size", type=int, default=32)
-    ap.add_argument("--seconds", type=float, default=10.0)
-    ap.add_argument("--num-actions", type=int, default=18)
-    args = ap.parse_args()
-    device = args.device
+def run(device=None, clients=8, batch_size=32, seconds=10.0, num_actions=18):
+    """Returns (total_requests, batch_sizes); used by tests and main()."""
+    device = device or ("cuda:0" if torch.cuda.is_available() else "cpu")
+
+    class args:
+        pass
+
+    args.clients, args.batch_size = clients, batch_size
+    args.seconds, args.num_actions = seconds, num_actions
 
     model = AtariNet(num_actions=args.num_actions).to(device)
     if device.startswith("cuda"):
@@ -90,6 +90,18 @@ def main():
         "(mean batch %.1f, max %d) on %s"
         % (total, dt, total / dt, len(batch_sizes), mean_b, max(batch_sizes or [0]), device)
     )
+    return total, batch_sizes
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--device", default=None)
+    ap.add_argument("--clients", type=int, default=8)
+    ap.add_argument("--batch-size", type=int, default=32)
+    ap.add_argument("--seconds", type=float, default=10.0)
+    ap.add_argument("--num-actions", type=int, default=18)
+    a = ap.parse_args()
+    run(a.device, a.clients, a.batch_size, a.seconds, a.num_actions)
 
 
 if __name__ == "__main__":

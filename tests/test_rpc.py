@@ -451,3 +451,23 @@ class TestSelfCall:
         r.define("boom", boom)
         with pytest.raises(moolib_amd.RpcError, match="kablam"):
             r.sync("me3", "boom")
+
+
+class TestInferenceServerExample:
+    @pytest.mark.timeout(120)
+    def test_dynamic_batching_actually_batches(self):
+        """examples/inference_server.py: with N concurrent callers and a
+        slow handler, the latency model must form multi-request batches."""
+        import os
+        import sys
+
+        sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", "examples"))
+        import inference_server
+
+        total, batch_sizes = inference_server.run(
+            device="cpu", clients=6, batch_size=32, seconds=6.0, num_actions=6
+        )
+        assert total > 50
+        mean_b = sum(batch_sizes) / len(batch_sizes)
+        assert mean_b > 1.3, mean_b  # singles-forever would be 1.0
+        assert max(batch_sizes) >= 3
