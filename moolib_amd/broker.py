@@ -26,9 +26,9 @@ def main():
     broker_rpc = moolib_amd.Rpc()
     broker_rpc.set_name("broker")
     broker = moolib_amd.Broker(broker_rpc)
-    broker_rpc.listen(flags.address)
+    bound = broker_rpc.listen(flags.address)
 
-    print("Broker listening at %s" % flags.address)
+    print("Broker listening at %s" % ", ".join(bound), flush=True)
 
     try:
         while True:
