@@ -91,7 +91,11 @@ void Accumulator::setup() {
       uint32_t nParams = r.u32();
       uint32_t nBuffers = r.u32();
       std::string statePayload(r.str());
-      if (nParams + nBuffers > f.tensors.size()) throw RpcError("model update tensor shortfall");
+      // 64-bit arithmetic: uint32 nParams + nBuffers can wrap on a corrupt
+      // frame, turning the bound check into out-of-bounds iterator math.
+      if (uint64_t(nParams) > f.tensors.size() || uint64_t(nBuffers) > f.tensors.size() ||
+          uint64_t(nParams) + uint64_t(nBuffers) > f.tensors.size())
+        throw RpcError("model update tensor shortfall");
       std::lock_guard<std::mutex> lk(self->mu_);
       self->pendingParams_.assign(f.tensors.begin(), f.tensors.begin() + nParams);
       self->pendingBuffers_.assign(f.tensors.begin() + nParams,

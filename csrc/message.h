@@ -29,7 +29,8 @@ enum class FrameKind : uint8_t {
   request = 2,       // rid/fid set, payload = serialized args
   response = 3,      // rid set, payload = serialized result
   errorResponse = 4, // rid set, payload = error string
-  ack = 5,           // rid set: "I received your request"
+  // (5 was a request-ack that was never part of the shipped protocol;
+  //  reliability is resend-on-reconnect + receiver dedupe + response acks)
   responseAck = 6,   // rid set: "I received your response"
   keepalive = 7,
   findPeer = 8,      // payload: peer name being looked for

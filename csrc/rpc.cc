@@ -527,6 +527,7 @@ void Rpc::handleRequest(ConnId id, Frame&& f) {
     IncomingKey key{peerUid, f.rid};
     auto [iit, inserted] = incoming_.try_emplace(key);
     iit->second.lastConn = id;
+    if (inserted) iit->second.created = now();
     if (!inserted) {
       if (iit->second.responded && !iit->second.acked) {
         engine_->send(id, iit->second.response);  // duplicate of a completed request
@@ -734,7 +735,18 @@ void Rpc::timerLoop() {
       // py-backed tensors; their destruction can take the GIL, so it must
       // happen outside mu_ (tombs vector, cleared after the lock scope).
       for (auto it = incoming_.begin(); it != incoming_.end();) {
-        if (it->second.responded && secondsSince(it->second.doneTime) > 60.0) {
+        bool reap = false;
+        if (it->second.responded) {
+          reap = secondsSince(it->second.doneTime) > 60.0;
+        } else {
+          // A deferred handler that never calls respond would otherwise pin
+          // the Incoming record (and its dedupe key) forever; the reference
+          // expires these via its timeout machinery (rpc.cc:1667-1760).
+          // A respond() after the reap is a silent no-op (record lookup
+          // fails), which matches a caller whose own timeout fired long ago.
+          reap = secondsSince(it->second.created) > 120.0;
+        }
+        if (reap) {
           tombs.push_back(std::move(it->second.response));
           it = incoming_.erase(it);
         } else {
@@ -746,7 +758,8 @@ void Rpc::timerLoop() {
         ConnId cid = it->first;
         ConnInfo& ci = it->second;
         double idle = secondsSince(ci.lastRecv);
-        if (ci.ready && idle > 2.0) {
+        if (ci.ready && idle > 2.0 && secondsSince(ci.lastKeepaliveSent) >= 2.0) {
+          ci.lastKeepaliveSent = t;
           Frame ka;
           ka.kind = FrameKind::keepalive;
           engine_->send(cid, std::move(ka));

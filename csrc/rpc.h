@@ -85,6 +85,7 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
     bool inbound = false;
     TimePoint lastRecv{};
     TimePoint established{};
+    TimePoint lastKeepaliveSent{};
   };
 
   // Per-address transport model (reference rpc.cc:640 "bandit" transport
@@ -139,6 +140,7 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
     bool acked = false;  // sender confirmed receipt: stored response freed
     Frame response;      // kept for re-send to duplicate requests until acked
     ConnId lastConn = 0;
+    TimePoint created{};   // for reaping handlers that never respond
     TimePoint doneTime{};
   };
 

@@ -217,7 +217,19 @@ class _BatchCollector:
                 deferred(out_i)
 
         t0 = time.monotonic()
-        self.process(tuple(batched_args), dict(batched_kwargs), respond_all, n)
+        try:
+            self.process(tuple(batched_args), dict(batched_kwargs), respond_all, n)
+        except Exception as e:  # noqa: BLE001
+            # Propagate the real handler exception to every caller in the
+            # batch; otherwise they all get the generic "deferred return
+            # dropped without a response" when the deferreds are destroyed.
+            msg = "%s: %s" % (type(e).__name__, e)
+            for deferred, _, _ in batch:
+                try:
+                    deferred.error(msg)
+                except Exception:  # noqa: BLE001 — already responded
+                    pass
+            return
         dt = time.monotonic() - t0
         with self.lock:
             self.proc_ema = dt if self.proc_ema is None else 0.8 * self.proc_ema + 0.2 * dt
@@ -245,11 +257,7 @@ class Rpc(_core.Rpc):
             return
 
         def process(bargs, bkwargs, respond_all, n):
-            try:
-                result = func(*bargs, **bkwargs)
-            except Exception as e:  # noqa: BLE001
-                raise e
-            respond_all(result)
+            respond_all(func(*bargs, **bkwargs))
 
         collector = _BatchCollector(batch_size, device, process, dynamic=dynamic_batching)
         self.define_deferred_raw(name, lambda d, *a, **kw: collector.add(d, a, kw))

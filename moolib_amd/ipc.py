@@ -33,7 +33,14 @@ from torch.multiprocessing import reductions
 # ref counter that only a consumer's rebuild/release would decrement, so the
 # same-process path must release it explicitly or every share pins its
 # storage forever (torch reductions.py does the same on its storage-cache
-# hit path). A TTL bounds never-consumed entries.
+# hit path). A TTL bounds the registry itself; TTL eviction deliberately
+# does NOT release the ref counter: the producer cannot know whether a
+# cross-process consumer already materialized the handle (that consumer's
+# rebuild adopted the counter and decrements it on free — releasing here
+# too would double-decrement and free HBM a consumer still aliases). The
+# cost is that a share that is never deserialized anywhere pins its
+# storage until the producer process exits; shares exist to be consumed,
+# so that leak is bounded and preferable to a GPU use-after-free.
 _local = {}  # key -> (tensor, deadline, func, args)
 _lock = threading.Lock()  # shares/materializes run on concurrent RPC threads
 _counter = [0]
@@ -54,10 +61,11 @@ def _release_producer_ref(func, args):
 
 
 def _evict_locked():
+    # Drops registry entries (and our strong tensor reference) only; see the
+    # module comment for why the IPC ref counter is never released here.
     now = time.monotonic()
-    dead = [k for k, ent in _local.items() if ent[1] < now]
-    evicted = [_local.pop(k) for k in dead]
-    return evicted
+    for k in [k for k, ent in _local.items() if ent[1] < now]:
+        _local.pop(k)
 
 
 def _materialize(pid, key, func, args):
@@ -83,12 +91,10 @@ class SharedCudaTensor:
     def __reduce__(self):
         func, args = reductions.reduce_tensor(self._tensor)
         with _lock:
-            evicted = _evict_locked()
+            _evict_locked()
             _counter[0] += 1
             key = _counter[0]
             _local[key] = (self._tensor, time.monotonic() + _TTL, func, args)
-        for ent in evicted:
-            _release_producer_ref(ent[2], ent[3])
         return (_materialize, (os.getpid(), key, func, args))
 
     def tensor(self):

@@ -229,6 +229,29 @@ class TestBatchedDefine:
             assert msg == "m"
             assert torch.allclose(s, tensors[i].sum())
 
+    def test_batched_handler_exception_propagates(self):
+        """Every caller in a batch gets the handler's real exception text,
+        not the generic dropped-deferred error."""
+        bs = 3
+        host = moolib_amd.Rpc()
+        host.set_name("host")
+        addr = host.listen("127.0.0.1:0")[0]
+
+        def fn(t):
+            raise ValueError("bad batch input %d" % t.shape[0])
+
+        host.define("f", fn, batch_size=bs)
+        clients, futures = [], []
+        for _ in range(bs):
+            c = moolib_amd.Rpc()
+            c.set_timeout(15)
+            c.connect(addr)
+            futures.append(c.async_("host", "f", torch.zeros(2)))
+            clients.append(c)
+        for f in futures:
+            with pytest.raises(Exception, match="bad batch input"):
+                f.result()
+
 
 class TestDynamicBatching:
     def test_partial_batch_flushes(self):
