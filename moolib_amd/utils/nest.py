@@ -1,7 +1,10 @@
 """Nested-structure helpers (map/flatten/zip over dict/list/tuple trees).
 
 Capability parity with the reference's examples/common/nest.py; used both by
-the library (batching) and by example code.
+the library (batching) and by example code. Implementation is our own:
+multi-nest operations flatten every nest to a parallel leaf list and rebuild
+along the first nest's structure, rather than threading iterators through a
+recursive map.
 """
 
 __all__ = ["map", "map_many", "flatten", "zip"]
@@ -38,23 +41,22 @@ def flatten(n):
         yield n
 
 
+def _rebuild(structure, leaves):
+    """Shape the flat iterator `leaves` like `structure` (inverse of flatten)."""
+    return map(lambda _: next(leaves), structure)
+
+
+def _leaf_groups(nests):
+    """Transpose several same-shaped nests into per-position leaf lists."""
+    columns = [list(flatten(n)) for n in nests]
+    return [list(group) for group in _builtin_zip(*columns)]
+
+
 def zip(*nests):
     """Zip leaves of several same-shaped nests into lists, shaped like nests[0]."""
-    first, *rest = nests
-    iters = [flatten(n) for n in rest]
-
-    def g(leaf):
-        return [leaf] + [next(i) for i in iters]
-
-    return map(g, first)
+    return _rebuild(nests[0], iter(_leaf_groups(nests)))
 
 
 def map_many(f, *nests):
     """Like zip, but applies f to each leaf-list."""
-    first, *rest = nests
-    iters = [flatten(n) for n in rest]
-
-    def g(leaf):
-        return f([leaf] + [next(i) for i in iters])
-
-    return map(g, first)
+    return _rebuild(nests[0], iter([f(g) for g in _leaf_groups(nests)]))

@@ -2,64 +2,85 @@
 
 Capability parity with the reference's examples/common/__init__.py:
 StatMean/StatSum, delta-based global stat allreduce with requeue-on-error
-(GlobalStatsAccumulator), RunningMeanStd.
+(GlobalStatsAccumulator), RunningMeanStd. The stat classes are our own
+design around one protocol every stat must speak so it can travel through
+the group allreduce: `+=` folds in a raw sample or a same-typed stat,
+`-` produces the delta not yet shipped cluster-wide, `result()` renders,
+`reset()` clears windowed state. Field names (`value`, `n`) are part of
+the pickle wire format between peers — do not rename.
 """
 import copy
-import dataclasses
 import logging
 
 import torch
 
 
-@dataclasses.dataclass
 class StatMean:
-    value: float = 0
-    n: int = 0
+    """Mean over everything folded in since the last reset.
 
-    def result(self):
-        if self.n == 0:
-            return None
-        return self.value / self.n
+    Stored as (numerator, count) so instances add and subtract exactly —
+    the delta protocol needs `a - b` then `c += delta` to be lossless.
+    """
 
-    def __sub__(self, other):
-        return StatMean(self.value - other.value, self.n - other.n)
+    __slots__ = ("value", "n")
 
-    def __iadd__(self, other):
-        if isinstance(other, StatMean):
-            self.value += other.value
-            self.n += other.n
-        else:
-            self.value += other
-            self.n += 1
+    def __init__(self, value=0, n=0):
+        self.value = value
+        self.n = n
+
+    def __iadd__(self, sample):
+        inc = (sample.value, sample.n) if isinstance(sample, StatMean) else (sample, 1)
+        self.value += inc[0]
+        self.n += inc[1]
         return self
 
+    def __sub__(self, baseline):
+        return StatMean(self.value - baseline.value, self.n - baseline.n)
+
+    def result(self):
+        return None if self.n == 0 else self.value / self.n
+
     def reset(self):
-        self.value = 0
-        self.n = 0
+        self.value, self.n = 0, 0
 
     def __repr__(self):
         return repr(self.result())
 
+    def __eq__(self, other):
+        if isinstance(other, StatMean):
+            return self.value == other.value and self.n == other.n
+        return NotImplemented
 
-@dataclasses.dataclass
+
 class StatSum:
-    value: float = 0
+    """Monotonic total; reset() is a no-op (lifetime counters survive
+    stat-window resets, matching the reference's semantics)."""
+
+    __slots__ = ("value",)
+
+    def __init__(self, value=0):
+        self.value = value
+
+    def __iadd__(self, sample):
+        self.value += sample.value if isinstance(sample, StatSum) else sample
+        return self
+
+    def __sub__(self, baseline):
+        return StatSum(self.value - baseline.value)
 
     def result(self):
         return self.value
-
-    def __sub__(self, other):
-        return StatSum(self.value - other.value)
-
-    def __iadd__(self, other):
-        self.value += other.value if isinstance(other, StatSum) else other
-        return self
 
     def reset(self):
         pass
 
     def __repr__(self):
         return repr(self.result())
+
+    def __eq__(self, other):
+        if isinstance(other, StatSum):
+            return self.value == other.value
+        return NotImplemented
 
 
 class GlobalStatsAccumulator:
