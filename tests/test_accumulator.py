@@ -412,9 +412,13 @@ class TestPeriodicModelBroadcast:
         leader_name = c.peers[0].acc.get_leader()
         leader = next(p for p in c.peers if p.rpc.get_name() == leader_name)
         follower = next(p for p in c.peers if p.rpc.get_name() != leader_name)
-        # any join-time state transfer settles first
-        c.pump(lambda: follower.state_got is not None or not follower.acc.has_new_state(),
-               deadline=5)
+        # The join-time user-state transfer always happens for a non-leader;
+        # wait for it POSITIVELY. (Waiting for "no state pending" instead is
+        # trivially true before the leader has even sent anything, and the
+        # late-arriving join transfer would then set state_got after we
+        # cleared it — the cross-test-order flake of round 1.)
+        ok = c.pump(lambda: follower.state_got is not None, deadline=10)
+        assert ok, "join-time state transfer never arrived"
         follower.state_got = None
         # desync the follower's weights; the periodic broadcast must repair
         with torch.no_grad():
