@@ -94,6 +94,17 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
+    if world > 1:
+        # 8-rank startup hardening: MIOpen's exhaustive conv find takes
+        # ~15 s/process and all ranks contend on one user find-db file
+        # lock; give each rank its own db and use the fast heuristic find.
+        # (N=1 keeps the default exhaustive find — same conditions as the
+        # single-GPU headline number.)
+        os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+        os.environ.setdefault(
+            "MIOPEN_USER_DB_PATH", "/tmp/miopen-rank%d" % rank
+        )
+
     # ---- 1. Fork env workers FIRST (clean pre-CUDA, pre-thread processes).
     import moolib_amd
 
@@ -205,8 +216,31 @@ def main():
             if ev == "optimize":
                 done += 1
 
+    def smi_telemetry(tag):
+        """Log GPU sclk/power/temp to stderr (bench-variance forensics:
+        r1 saw a 47-63k f/s box-to-box spread for identical code)."""
+        if not use_cuda or not os.environ.get("MOOLIB_AMD_BENCH_TELEMETRY"):
+            return
+        import subprocess
+
+        try:
+            out = subprocess.run(
+                ["rocm-smi", "--showgpuclocks", "--showpower", "--showtemp"],
+                capture_output=True, text=True, timeout=10,
+            ).stdout
+            keep = [
+                l for l in out.splitlines()
+                if any(k in l for k in ("sclk", "Power", "Temperature", "edge"))
+            ]
+            print("[telemetry %s rank %d]\n%s" % (tag, rank, "\n".join(keep)),
+                  file=sys.stderr, flush=True)
+        except Exception as e:  # noqa: BLE001 — telemetry must never kill the bench
+            print("[telemetry %s failed: %s]" % (tag, e), file=sys.stderr)
+
     # ---- warmup ----
+    smi_telemetry("pre-warmup")
     run_steps(args.warmup, args.max_seconds / 2)
+    smi_telemetry("post-warmup")
     if args.breakdown:
         peer.profile = True
         peer.phase_times = {}
@@ -219,6 +253,7 @@ def main():
     run_steps(args.steps, args.max_seconds)
     barrier_sync()
     elapsed = time.time() - t0
+    smi_telemetry("post-timed")
     frames = (vbs_stat.value - v0) * args.unroll_length  # global frames consumed
 
     if dist is not None:
