@@ -10,8 +10,10 @@
 //
 // MI355X note: targets live on the learner GPU; copies are issued
 // non-blocking so H2D of actor rollouts overlaps the learner's compute
-// stream. (The fused gather path in moolib_amd._kernels replaces per-slice
-// copies for the hot [T,B,84,84,4] frames case.)
+// stream. When moolib_amd._kernels is importable and a GPU is present,
+// the package registers its batched_copy kernel here (setFusedCopy) and
+// every stack()/cat() call moves ALL its leaves in one launch instead of
+// one ramp-dominated runtime copy per leaf.
 #include "batcher.h"
 
 #include <torch/csrc/autograd/python_variable.h>
@@ -67,7 +69,18 @@ void zipNest(py::handle a, py::handle b, const std::function<void(py::handle, py
 
 bool isTensor(py::handle h) { return THPVariable_Check(h.ptr()); }
 
+// Fused copy hook: fn(list[Tensor] dsts, list[Tensor] srcs) performing
+// dst.copy_(src) for every pair (one kernel launch for the CUDA pairs).
+// Leaked pointer: destroying a py::object at static-destruction time
+// would race interpreter teardown.
+py::object* g_fusedCopy = nullptr;
+
 }  // namespace
+
+void setBatcherFusedCopy(py::object fn) {
+  if (g_fusedCopy == nullptr) g_fusedCopy = new py::object();
+  *g_fusedCopy = fn.is_none() ? py::object() : std::move(fn);
+}
 
 Batcher::Batcher(int64_t size, py::object device, int64_t dim)
     : size_(size), dim_(dim) {
@@ -111,12 +124,21 @@ void Batcher::stack(py::object nest) {
     });
     fill_ = 0;
   }
+  bool fused = g_fusedCopy != nullptr && *g_fusedCopy;
+  py::list fdsts, fsrcs;
   zipNest(current_, nest, [&](py::handle t, py::handle s) {
     if (!isTensor(t) || !isTensor(s)) return;
     at::Tensor dst = THPVariable_Unpack(t.ptr());
     at::Tensor src = THPVariable_Unpack(s.ptr());
-    dst.select(dim_, fill_).copy_(src, /*non_blocking=*/true);
+    at::Tensor slot = dst.select(dim_, fill_);
+    if (fused) {
+      fdsts.append(py::reinterpret_steal<py::object>(THPVariable_Wrap(slot)));
+      fsrcs.append(py::reinterpret_borrow<py::object>(s));
+    } else {
+      slot.copy_(src, /*non_blocking=*/true);
+    }
   });
+  if (fused && py::len(fdsts) > 0) (*g_fusedCopy)(fdsts, fsrcs);
   ++fill_;
   if (fill_ >= size_) completeBatch();
 }
@@ -143,12 +165,22 @@ void Batcher::cat(py::object nest) {
       fill_ = 0;
     }
     int64_t n = std::min(srcLen - srcOff, size_ - fill_);
+    bool fused = g_fusedCopy != nullptr && *g_fusedCopy;
+    py::list fdsts, fsrcs;
     zipNest(current_, nest, [&](py::handle t, py::handle s) {
       if (!isTensor(t) || !isTensor(s)) return;
       at::Tensor dst = THPVariable_Unpack(t.ptr());
       at::Tensor src = THPVariable_Unpack(s.ptr());
-      dst.narrow(dim_, fill_, n).copy_(src.narrow(dim_, srcOff, n), /*non_blocking=*/true);
+      at::Tensor dslot = dst.narrow(dim_, fill_, n);
+      at::Tensor sslot = src.narrow(dim_, srcOff, n);
+      if (fused) {
+        fdsts.append(py::reinterpret_steal<py::object>(THPVariable_Wrap(dslot)));
+        fsrcs.append(py::reinterpret_steal<py::object>(THPVariable_Wrap(sslot)));
+      } else {
+        dslot.copy_(sslot, /*non_blocking=*/true);
+      }
     });
+    if (fused && py::len(fdsts) > 0) (*g_fusedCopy)(fdsts, fsrcs);
     fill_ += n;
     srcOff += n;
     if (fill_ >= size_) completeBatch();

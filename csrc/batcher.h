@@ -10,6 +10,10 @@
 
 namespace mrl {
 
+// Register fn(list[Tensor] dsts, list[Tensor] srcs) as the fused
+// multi-leaf copy (moolib_amd._kernels.batched_copy); None unregisters.
+void setBatcherFusedCopy(py::object fn);
+
 class Batcher {
  public:
   Batcher(int64_t size, py::object device, int64_t dim);

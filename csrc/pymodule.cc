@@ -605,6 +605,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("size", &Batcher::size)
       .def("get", &Batcher::get)
       .def("_pop_future", &Batcher::popFuture);
+  m.def("_set_batcher_fused_copy", &setBatcherFusedCopy,
+        "register the _kernels.batched_copy fused slice-copy hook");
 
   py::class_<AccumulatorWrapper>(m, "Accumulator")
       .def(py::init<const std::string&, py::object, py::object, py::object>(), py::arg("name"),

@@ -694,9 +694,163 @@ at::Tensor bias_add2_fwd(at::Tensor x, at::Tensor b1, at::Tensor s,
   return out;
 }
 
+// ------------------------------------------------ batched slice copies
+//
+// The Batcher (csrc/batcher.cc; reference semantics src/moolib.cc:651-752)
+// fills preallocated [T,B,...] targets with one copy_ per nest leaf per
+// call — on the IMPALA hot path that is ~8 tiny __amd_rocclr_copyBuffer
+// launches per rollout step (~0.7 ms/optimizer step of ramp-dominated
+// copies in the r1 profile). This kernel moves every leaf of one
+// stack()/cat() call in a single launch: each entry is a strided
+// rows x rowBytes region, the descriptor table travels in the kernel-arg
+// block (no H2D staging), and blockIdx.y picks the entry so the per-entry
+// vector width is wave-uniform.
+
+struct CopyDesc {
+  const uint8_t* src;
+  uint8_t* dst;
+  int64_t rows;
+  int64_t rowBytes;
+  int64_t srcStrideB;  // bytes between rows
+  int64_t dstStrideB;
+  int32_t vec;  // 16 / 4 / 1: granularity valid for ptrs, strides, rowBytes
+  int32_t pad;
+};
+
+constexpr int kMaxCopyDescs = 24;  // 24 * 56 B comfortably under the 4 KB arg limit
+struct CopyDescPack {
+  CopyDesc d[kMaxCopyDescs];
+  int32_t n;
+};
+
+template <typename V>
+DEV_INLINE void copyRun(const CopyDesc& e, int64_t start, int64_t step, int64_t total) {
+  for (int64_t idx = start; idx < total; idx += step) {
+    int64_t byte = idx * (int64_t)sizeof(V);
+    int64_t row = byte / e.rowBytes;
+    int64_t col = byte - row * e.rowBytes;
+    *reinterpret_cast<V*>(e.dst + row * e.dstStrideB + col) =
+        *reinterpret_cast<const V*>(e.src + row * e.srcStrideB + col);
+  }
+}
+
+__global__ void batched_copy_kernel(CopyDescPack pack) {
+  const CopyDesc& e = pack.d[blockIdx.y];
+  int64_t units = e.rows * e.rowBytes / e.vec;
+  int64_t start = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t step = (int64_t)gridDim.x * blockDim.x;
+  if (e.vec == 16) copyRun<uint4>(e, start, step, units);
+  else if (e.vec == 4) copyRun<uint32_t>(e, start, step, units);
+  else copyRun<uint8_t>(e, start, step, units);
+}
+
+namespace {
+
+// Express a tensor view as rows x contiguous-rowElems with a uniform row
+// stride (the shape select()/narrow() of a contiguous buffer produces).
+// Returns false for layouts that need a general copy.
+bool rowDecompose(const at::Tensor& t, int64_t& rows, int64_t& rowElems, int64_t& strideEl) {
+  int64_t inner = 1;
+  int d = t.dim() - 1;
+  for (; d >= 0; --d) {
+    if (t.size(d) == 1) continue;
+    if (t.stride(d) != inner) break;
+    inner *= t.size(d);
+  }
+  rowElems = inner;
+  if (d < 0) {  // fully contiguous
+    rows = 1;
+    strideEl = inner;
+    return true;
+  }
+  rows = t.size(d);
+  strideEl = t.stride(d);
+  for (int i = d - 1; i >= 0; --i) {
+    if (t.size(i) == 1) continue;
+    if (t.stride(i) != strideEl * rows) return false;
+    rows *= t.size(i);
+  }
+  return strideEl >= rowElems;
+}
+
+int vecWidthFor(const void* a, const void* b, int64_t rowBytes, int64_t sA, int64_t sB) {
+  auto ok = [&](int64_t v) {
+    return ((uintptr_t)a % v == 0) && ((uintptr_t)b % v == 0) && (rowBytes % v == 0) &&
+           (sA % v == 0) && (sB % v == 0);
+  };
+  if (ok(16)) return 16;
+  if (ok(4)) return 4;
+  return 1;
+}
+
+}  // namespace
+
+// dsts[i] <- srcs[i] for same-shaped same-dtype CUDA pairs, all in one
+// launch (chunked by kMaxCopyDescs). Pairs this kernel can't express
+// (cross-device, CPU, exotic strides) fall back to copy_ individually, so
+// the call is semantically total.
+void batched_copy(std::vector<at::Tensor> dsts, std::vector<at::Tensor> srcs) {
+  TORCH_CHECK(dsts.size() == srcs.size(), "batched_copy: length mismatch");
+  CopyDescPack pack;
+  pack.n = 0;
+  int64_t maxUnits = 0;
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  auto flush = [&]() {
+    if (pack.n == 0) return;
+    int threads = 256;
+    int64_t blocksX = std::min<int64_t>((maxUnits + threads - 1) / threads, 1024);
+    hipLaunchKernelGGL(batched_copy_kernel, dim3(blocksX, pack.n), dim3(threads), 0, stream,
+                       pack);
+    pack.n = 0;
+    maxUnits = 0;
+  };
+  for (size_t i = 0; i < dsts.size(); ++i) {
+    at::Tensor& dst = dsts[i];
+    at::Tensor& src = srcs[i];
+    bool fused = false;
+    if (dst.is_cuda() && src.is_cuda() && dst.device() == src.device() &&
+        dst.scalar_type() == src.scalar_type() && dst.numel() == src.numel() &&
+        dst.numel() > 0) {
+      int64_t dRows, dRowEl, dStride, sRows, sRowEl, sStride;
+      if (rowDecompose(dst, dRows, dRowEl, dStride) &&
+          rowDecompose(src, sRows, sRowEl, sStride)) {
+        // Unify row shapes: a fully contiguous side adopts the other's rows.
+        if (dRows == 1 && sRows > 1) {
+          dRows = sRows;
+          dRowEl = sRowEl;
+          dStride = sRowEl;
+        } else if (sRows == 1 && dRows > 1) {
+          sRows = dRows;
+          sRowEl = dRowEl;
+          sStride = dRowEl;
+        }
+        if (dRows == sRows && dRowEl == sRowEl) {
+          int64_t esize = dst.element_size();
+          CopyDesc e;
+          e.src = (const uint8_t*)src.data_ptr();
+          e.dst = (uint8_t*)dst.data_ptr();
+          e.rows = dRows;
+          e.rowBytes = dRowEl * esize;
+          e.srcStrideB = sStride * esize;
+          e.dstStrideB = dStride * esize;
+          e.vec = vecWidthFor(e.src, e.dst, e.rowBytes, e.srcStrideB, e.dstStrideB);
+          pack.d[pack.n++] = e;
+          maxUnits = std::max(maxUnits, e.rows * e.rowBytes / e.vec);
+          if (pack.n == kMaxCopyDescs) flush();
+          fused = true;
+        }
+      }
+    }
+    if (!fused) dst.copy_(src, /*non_blocking=*/true);
+  }
+  flush();
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "moolib_amd gfx950 HIP kernels";
   m.def("register_host_memory", &register_host_memory);
+  m.def("batched_copy", &batched_copy,
+        "copy many (dst<-src) slice pairs in one kernel launch (gfx950)");
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd, "NHWC 3x3/2 maxpool forward (gfx950)");
   m.def("maxpool3x3s2_bwd", &maxpool3x3s2_bwd, "NHWC 3x3/2 maxpool backward (gather, no atomics)");
   m.def("bias_relu_fwd", &bias_relu_fwd, "fused conv-bias + relu, one NHWC pass (gfx950)");

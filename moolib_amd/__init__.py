@@ -98,6 +98,13 @@ Queue.get = _queue_get
 _core.Batcher.__await__ = _queue_await
 _core.Batcher.__iter__ = _queue_await
 
+# Fused batcher copies: with a GPU present, every Batcher.stack/cat call
+# moves all its tensor leaves in ONE _kernels.batched_copy launch instead
+# of one runtime copy per leaf (the kernel falls back to copy_ per pair
+# for layouts/devices it cannot express, so semantics are unchanged).
+if _kernels is not None and torch.cuda.is_available():
+    _core._set_batcher_fused_copy(_kernels.batched_copy)
+
 
 # Rpc with batched defines --------------------------------------------------
 
