@@ -300,7 +300,91 @@ try:
 except ImportError:  # pragma: no cover — during staged bring-up
     pass
 
-from ._core import EnvPool, EnvStepperFuture  # noqa: F401
+from ._core import EnvStepperFuture  # noqa: F401
+from ._core import EnvPool as _EnvPoolCore
+
+
+class _DispatchedStep:
+    """Future for a step whose CUDA action is still in flight to the CPU.
+
+    result() waits for the dispatcher to hand the action to the workers
+    (i.e. the real _core future to exist), then delegates."""
+
+    __slots__ = ("_event", "_holder")
+
+    def __init__(self):
+        self._event = threading.Event()
+        self._holder = []
+
+    def _resolve(self, fut_or_exc):
+        self._holder.append(fut_or_exc)
+        self._event.set()
+
+    def result(self):
+        self._event.wait()
+        v = self._holder[0]
+        if isinstance(v, BaseException):
+            raise v
+        return v.result()
+
+
+class EnvPool(_EnvPoolCore):
+    """EnvPool with asynchronous CUDA action staging.
+
+    A CUDA action tensor (the actor forward's output, usually still being
+    computed when step() is called) is copied to a persistent pinned slot
+    non_blocking and handed to a dispatcher thread that waits on a HIP
+    event before waking the workers — the act loop never blocks on the
+    device (the reference stages through a pinned tensor but then does a
+    synchronous stream sync inline, src/env.cc:309-319). CPU actions take
+    the direct path unchanged.
+    """
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self._staging = {}  # batch_index -> [pinned, torch.cuda.Event]
+        self._dispatch_q = None
+        self._dispatcher = None
+
+    def _ensure_dispatcher(self):
+        if self._dispatcher is None:
+            import queue as _queue
+
+            self._dispatch_q = _queue.SimpleQueue()
+            self._dispatcher = threading.Thread(
+                target=self._dispatch_loop, daemon=True, name="envpool-dispatch"
+            )
+            self._dispatcher.start()
+
+    def _dispatch_loop(self):
+        while True:
+            item = self._dispatch_q.get()
+            if item is None:
+                return
+            b, pinned, ev, lazy = item
+            try:
+                ev.synchronize()  # D2H copy into the pinned slot has landed
+                lazy._resolve(_EnvPoolCore.step(self, b, pinned))
+            except BaseException as e:  # noqa: BLE001 — surfaced via result()
+                lazy._resolve(e)
+
+    def step(self, batch_index, action):
+        if not (isinstance(action, torch.Tensor) and action.is_cuda):
+            return super().step(batch_index, action)
+        self._ensure_dispatcher()
+        slot = self._staging.get(batch_index)
+        if slot is None:
+            pinned = torch.empty(
+                action.shape, dtype=torch.int64, device="cpu"
+            ).pin_memory()
+            slot = [pinned, torch.cuda.Event()]
+            self._staging[batch_index] = slot
+        pinned, ev = slot
+        pinned.copy_(action.detach().to(torch.int64), non_blocking=True)
+        ev.record()
+        lazy = _DispatchedStep()
+        self._dispatch_q.put((batch_index, pinned, ev, lazy))
+        return lazy
 
 
 class EnvStepper:

@@ -610,3 +610,30 @@ class TestBatchedCopy:
         c.cat(src)  # 6 fill + 4 carry
         out = c.get()
         assert torch.equal(out["v"], src["v"][:6])
+
+
+@gpu
+@requires_gpu
+class TestAsyncActionStaging:
+    def test_cuda_action_roundtrip(self):
+        """step() with a CUDA action must not block on the device in the
+        caller and must deliver the same actions to the workers."""
+        from moolib_amd.envs import SyntheticAtariEnv
+
+        pool = moolib_amd.EnvPool(
+            lambda: SyntheticAtariEnv(num_actions=6),
+            num_processes=2,
+            batch_size=8,
+            num_batches=2,
+        )
+        try:
+            # prime both batches (CPU path)
+            for b in range(2):
+                pool.step(b, torch.zeros(8, dtype=torch.int64)).result()
+            for i in range(5):
+                act = torch.randint(0, 6, (8,), device="cuda")
+                fut = pool.step(i % 2, act)
+                out = fut.result()
+                assert out["state"].shape[0] == 8
+        finally:
+            del pool
