@@ -55,11 +55,12 @@ py::object rpcErrorType() { return *g_rpcErrorType; }
 class RpcDeferredReturn {
  public:
   RpcDeferredReturn() = default;
-  explicit RpcDeferredReturn(RespondFn respond) : respond_(std::move(respond)) {}
+  explicit RpcDeferredReturn(RespondFn respond, bool ipcLocal = false)
+      : respond_(std::move(respond)), ipcLocal_(ipcLocal) {}
   RpcDeferredReturn(const RpcDeferredReturn&) = delete;
   RpcDeferredReturn& operator=(const RpcDeferredReturn&) = delete;
   RpcDeferredReturn(RpcDeferredReturn&& o) noexcept
-      : respond_(std::move(o.respond_)), called_(o.called_) {
+      : respond_(std::move(o.respond_)), called_(o.called_), ipcLocal_(o.ipcLocal_) {
     o.respond_ = nullptr;
     o.called_ = true;
   }
@@ -73,7 +74,7 @@ class RpcDeferredReturn {
     if (!respond_ || called_) throw RpcError("deferred return already used");
     called_ = true;
     std::vector<at::Tensor> tensors;
-    std::string payload = serializeObject(value, tensors);
+    std::string payload = serializeObject(value, tensors, ipcLocal_);
     auto r = std::move(respond_);
     py::gil_scoped_release rel;
     r(std::move(payload), std::move(tensors), false);
@@ -89,6 +90,7 @@ class RpcDeferredReturn {
  private:
   RespondFn respond_;
   bool called_ = false;
+  bool ipcLocal_ = false;  // requester is same-machine: CUDA replies go as hipIpc handles
 };
 
 // ----------------------------------------------------------------- Rpc
@@ -135,8 +137,11 @@ class RpcWrapper {
 
   void define(const std::string& name, py::function fn) {
     auto g = std::make_shared<PyGuard>(fn);
-    rpc_->define(name, [g](Frame f, const std::string& from, RespondFn respond) {
+    std::weak_ptr<Rpc> wr = rpc_;
+    rpc_->define(name, [g, wr](Frame f, const std::string& from, RespondFn respond) {
       if (!pyAlive()) return;
+      auto rpc = wr.lock();
+      bool local = rpc && rpc->peerIsLocal(from);
       std::string payload;
       std::vector<at::Tensor> tensors;
       bool isErr = false;
@@ -145,7 +150,7 @@ class RpcWrapper {
         try {
           auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
           py::object r = g->obj(*args, **kwargs);
-          payload = serializeObject(r, tensors);
+          payload = serializeObject(r, tensors, local);
         } catch (py::error_already_set& e) {
           payload = e.what();
           isErr = true;
@@ -160,12 +165,15 @@ class RpcWrapper {
 
   void defineDeferred(const std::string& name, py::function fn) {
     auto g = std::make_shared<PyGuard>(fn);
-    rpc_->define(name, [g](Frame f, const std::string& from, RespondFn respond) {
+    std::weak_ptr<Rpc> wr = rpc_;
+    rpc_->define(name, [g, wr](Frame f, const std::string& from, RespondFn respond) {
       if (!pyAlive()) return;
+      auto rpc = wr.lock();
+      bool local = rpc && rpc->peerIsLocal(from);
       py::gil_scoped_acquire gil;
       try {
         auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
-        py::object deferred = py::cast(RpcDeferredReturn(std::move(respond)));
+        py::object deferred = py::cast(RpcDeferredReturn(std::move(respond), local));
         g->obj(deferred, *args, **kwargs);
       } catch (py::error_already_set& e) {
         // The deferred return (if not moved from) responds with an error when
@@ -179,12 +187,15 @@ class RpcWrapper {
 
   PyQueue defineQueue(const std::string& name) {
     PyQueue q;
-    rpc_->define(name, [q](Frame f, const std::string& from, RespondFn respond) mutable {
+    std::weak_ptr<Rpc> wr = rpc_;
+    rpc_->define(name, [q, wr](Frame f, const std::string& from, RespondFn respond) mutable {
       if (!pyAlive()) return;
+      auto rpc = wr.lock();
+      bool local = rpc && rpc->peerIsLocal(from);
       py::gil_scoped_acquire gil;
       try {
         auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
-        py::object deferred = py::cast(RpcDeferredReturn(std::move(respond)));
+        py::object deferred = py::cast(RpcDeferredReturn(std::move(respond), local));
         q.enqueue(py::make_tuple(deferred, args, kwargs));
       } catch (py::error_already_set& e) {
         MRL_LOG_ERROR("queue handler failed: %s", e.what());
@@ -200,7 +211,7 @@ class RpcWrapper {
   PyFuture asyncCall(const std::string& peer, const std::string& func, py::args args,
                      py::kwargs kwargs) {
     std::vector<at::Tensor> tensors;
-    std::string payload = serializeCall(args, kwargs, tensors);
+    std::string payload = serializeCall(args, kwargs, tensors, rpc_->peerIsLocal(peer));
     PyFuture fut;
     auto st = fut.state();
     py::gil_scoped_release rel;
@@ -227,7 +238,7 @@ class RpcWrapper {
   void asyncCallback(const std::string& peer, const std::string& func, py::function cb,
                      py::args args, py::kwargs kwargs) {
     std::vector<at::Tensor> tensors;
-    std::string payload = serializeCall(args, kwargs, tensors);
+    std::string payload = serializeCall(args, kwargs, tensors, rpc_->peerIsLocal(peer));
     auto g = std::make_shared<PyGuard>(cb);
     py::gil_scoped_release rel;
     rpc_->sendRequest(peer, func, std::move(payload), std::move(tensors),

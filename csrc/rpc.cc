@@ -155,6 +155,13 @@ std::vector<std::string> Rpc::connectedPeers() {
   return out;
 }
 
+bool Rpc::peerIsLocal(const std::string& peerName) {
+  if (peerName == name_) return true;  // self-calls dispatch locally
+  std::lock_guard<std::mutex> lk(mu_);
+  auto it = peers_.find(peerName);
+  return it != peers_.end() && !it->second.machine.empty() && it->second.machine == machineId_;
+}
+
 // ----------------------------------------------------------- greeting
 
 void Rpc::sendGreeting(ConnId id) {
@@ -266,6 +273,7 @@ void Rpc::handleGreeting(ConnId id, Frame& f) {
       p.addrs.clear();
     }
     p.uid = peerUid;
+    p.machine = peerMachine;
     for (auto& a : addrs) {
       // Unix addrs are only usable from the same machine.
       if (a.rfind("unix://", 0) == 0 && a.find(machineId_) == std::string::npos) continue;

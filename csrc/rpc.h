@@ -70,6 +70,11 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
   // Names of peers we currently have a live, greeted connection to.
   std::vector<std::string> connectedPeers();
 
+  // True if `peerName` greeted us from this machine (machineId match) —
+  // the gate for hipIpc zero-copy GPU-tensor serialization. Unknown or
+  // not-yet-greeted peers report false (callers fall back to staging).
+  bool peerIsLocal(const std::string& peerName);
+
   void shutdown();
   bool isShutdown() const { return stopping_.load(); }
 
@@ -100,6 +105,7 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
 
   struct PeerInfo {
     std::string uid;
+    std::string machine;             // machineId from the greeting
     std::vector<std::string> addrs;  // candidate addresses
     ConnId activeConn = 0;           // ready connection
     ConnId connecting = 0;           // outbound connect in flight

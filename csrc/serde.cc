@@ -45,9 +45,27 @@ bool isNumpyArray(py::handle h) {
   return ndarrayType->ptr() && py::isinstance(h, *ndarrayType);
 }
 
+// moolib_amd.ipc.share — wraps a CUDA tensor so pickling it produces a
+// hipIpc handle (torch multiprocessing reductions). Resolved lazily and
+// cached; empty if the module is unavailable or MOOLIB_AMD_NO_IPC_RPC is
+// set (the staging path then applies).
+py::object& ipcShare() {
+  static py::object* f = []() -> py::object* {
+    if (std::getenv("MOOLIB_AMD_NO_IPC_RPC")) return new py::object();
+    try {
+      return new py::object(py::module_::import("moolib_amd.ipc").attr("share"));
+    } catch (...) {
+      PyErr_Clear();
+      return new py::object();
+    }
+  }();
+  return *f;
+}
+
 }  // namespace
 
-void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors) {
+void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors,
+                 bool ipcLocal) {
   if (obj.is_none()) {
     w.u8(tNone);
   } else if (py::isinstance<py::bool_>(obj)) {
@@ -78,6 +96,22 @@ void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors
     w.str(std::string_view(PyBytes_AS_STRING(obj.ptr()), PyBytes_GET_SIZE(obj.ptr())));
   } else if (THPVariable_Check(obj.ptr())) {
     at::Tensor t = THPVariable_Unpack(obj.ptr());
+    if (!t.device().is_cpu() && ipcLocal && ipcShare()) {
+      // Same-machine destination: ship a hipIpc handle, not the bytes —
+      // the receiver materializes a tensor aliasing this process's HBM.
+      // (The reference fatals on CUDA tensors over the wire,
+      // src/rpc.cc:661-667; our cross-node path stages through the CPU.)
+      try {
+        py::object shared =
+            ipcShare()(py::reinterpret_steal<py::object>(THPVariable_Wrap(t.detach())));
+        w.u8(tPickle);
+        py::bytes b = pickleDumps()(shared, 2);
+        w.str(std::string_view(PyBytes_AS_STRING(b.ptr()), PyBytes_GET_SIZE(b.ptr())));
+        return;
+      } catch (...) {
+        PyErr_Clear();  // e.g. unshareable storage — fall back to staging
+      }
+    }
     if (t.device().is_cpu()) {
       w.u8(tTensor);
     } else {
@@ -112,19 +146,19 @@ void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors
     auto t = py::reinterpret_borrow<py::tuple>(obj);
     w.u8(tTuple);
     w.u32(static_cast<uint32_t>(t.size()));
-    for (auto item : t) serializePy(item, w, tensors);
+    for (auto item : t) serializePy(item, w, tensors, ipcLocal);
   } else if (py::isinstance<py::list>(obj)) {
     auto l = py::reinterpret_borrow<py::list>(obj);
     w.u8(tList);
     w.u32(static_cast<uint32_t>(l.size()));
-    for (auto item : l) serializePy(item, w, tensors);
+    for (auto item : l) serializePy(item, w, tensors, ipcLocal);
   } else if (py::isinstance<py::dict>(obj)) {
     auto d = py::reinterpret_borrow<py::dict>(obj);
     w.u8(tDict);
     w.u32(static_cast<uint32_t>(d.size()));
     for (auto item : d) {
-      serializePy(item.first, w, tensors);
-      serializePy(item.second, w, tensors);
+      serializePy(item.first, w, tensors, ipcLocal);
+      serializePy(item.second, w, tensors, ipcLocal);
     }
   } else {
     w.u8(tPickle);
@@ -202,10 +236,11 @@ py::object deserializePy(WireReader& r, const std::vector<at::Tensor>& tensors) 
   }
 }
 
-std::string serializeCall(py::tuple args, py::dict kwargs, std::vector<at::Tensor>& tensors) {
+std::string serializeCall(py::tuple args, py::dict kwargs, std::vector<at::Tensor>& tensors,
+                          bool ipcLocal) {
   WireWriter w;
-  serializePy(args, w, tensors);
-  serializePy(kwargs, w, tensors);
+  serializePy(args, w, tensors, ipcLocal);
+  serializePy(kwargs, w, tensors, ipcLocal);
   return std::move(w.out);
 }
 
@@ -217,9 +252,9 @@ std::pair<py::tuple, py::dict> deserializeCall(std::string_view payload,
   return {py::reinterpret_borrow<py::tuple>(args), py::reinterpret_borrow<py::dict>(kwargs)};
 }
 
-std::string serializeObject(py::handle obj, std::vector<at::Tensor>& tensors) {
+std::string serializeObject(py::handle obj, std::vector<at::Tensor>& tensors, bool ipcLocal) {
   WireWriter w;
-  serializePy(obj, w, tensors);
+  serializePy(obj, w, tensors, ipcLocal);
   return std::move(w.out);
 }
 

@@ -18,15 +18,22 @@ namespace mrl {
 
 namespace py = pybind11;
 
-void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors);
+// ipcLocal: destination peer is on this machine — CUDA tensors serialize
+// as hipIpc handles (moolib_amd.ipc.share) instead of being staged through
+// the CPU; the receiver materializes a tensor aliasing the sender's HBM
+// (zero copies). False (default) keeps the CPU-staging wire path.
+void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors,
+                 bool ipcLocal = false);
 py::object deserializePy(WireReader& r, const std::vector<at::Tensor>& tensors);
 
 // Convenience: serialize (args, kwargs) into payload+tensors and back.
-std::string serializeCall(py::tuple args, py::dict kwargs, std::vector<at::Tensor>& tensors);
+std::string serializeCall(py::tuple args, py::dict kwargs, std::vector<at::Tensor>& tensors,
+                          bool ipcLocal = false);
 std::pair<py::tuple, py::dict> deserializeCall(std::string_view payload,
                                                const std::vector<at::Tensor>& tensors);
 
-std::string serializeObject(py::handle obj, std::vector<at::Tensor>& tensors);
+std::string serializeObject(py::handle obj, std::vector<at::Tensor>& tensors,
+                            bool ipcLocal = false);
 py::object deserializeObject(std::string_view payload, const std::vector<at::Tensor>& tensors);
 
 }  // namespace mrl
