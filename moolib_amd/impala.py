@@ -436,9 +436,7 @@ class ImpalaPeer:
         # graph replays read the updated weights; capturable ops only.
         from moolib_amd.ops import conv3x3 as _c3
 
-        import os as _os
-
-        if _os.environ.get("MOOLIB_AMD_CONV3_KERNEL"):
+        if self.is_cuda and _c3.available(16, 16):
             _c3.repack(self.fwd_model)
         return norm
 

@@ -27,12 +27,13 @@ def _k():
 
 
 def available(C, K):
-    # Opt-in: measured on MI355X the v1 kernel (16-pixel M-tiles, no LDS
-    # tap reuse) is ~20% slower than MIOpen's tuned igemm on the actor path
-    # (44.4k vs 50.5k f/s same-box A/B) despite the fusion wins. Numerics
-    # are validated (tests/test_gpu.py::TestConv3x3Fused); revisit with
-    # LDS-staged input tiles + larger M-tiles before defaulting on.
-    if not os.environ.get("MOOLIB_AMD_CONV3_KERNEL"):
+    # Default ON for the actor path since r2: with packed weights cached in
+    # version-checked persistent buffers (repack() once per optimizer step;
+    # the r1 regression was per-replay repacking inside the captured
+    # forward), same-box interleaved A/B measured the kernel path ~4%
+    # faster end-to-end (gpurun_out/r2_{on,off}*.json: 60.7k vs 58.4k and
+    # 58.7k vs 56.2k f/s). MOOLIB_AMD_CONV3_KERNEL=0 reverts to MIOpen.
+    if os.environ.get("MOOLIB_AMD_CONV3_KERNEL", "1") == "0":
         return False
     if C not in (16, 32) or K not in (16, 32):
         return False

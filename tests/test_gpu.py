@@ -534,7 +534,7 @@ class TestConv3x3Fused:
         try:
             with torch.no_grad():
                 out_f, _ = model(inputs, tuple())
-                del os.environ["MOOLIB_AMD_CONV3_KERNEL"]
+                os.environ["MOOLIB_AMD_CONV3_KERNEL"] = "0"  # force MIOpen path
                 out_e, _ = model(inputs, tuple())
         finally:
             os.environ.pop("MOOLIB_AMD_CONV3_KERNEL", None)
