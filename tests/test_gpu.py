@@ -637,3 +637,75 @@ class TestAsyncActionStaging:
                 assert out["state"].shape[0] == 8
         finally:
             del pool
+
+
+@gpu
+@requires_gpu
+class TestBf16LearningParity:
+    def test_gradient_direction_bf16_vs_fp32(self):
+        """The shipped precision mode (bf16 weights + fp32 V-trace/loss)
+        must produce gradients pointing the same way as a full-fp32 pass
+        through the SAME learn pipeline (end-to-end
+        bf16 evidence, not just per-kernel tolerances)."""
+        import copy
+
+        from moolib_amd.models.atari import AtariNet
+        from moolib_amd.ops import vtrace
+        from moolib_amd.ops import fused_loss
+
+        torch.manual_seed(7)
+        T, B, A = 10, 8, 6
+        m32 = AtariNet(num_actions=A).to("cuda")
+        m16 = copy.deepcopy(m32).to(torch.bfloat16)
+        m16 = m16.to(memory_format=torch.channels_last)
+
+        batch = {
+            "state": torch.randint(0, 255, (T + 1, B, 4, 84, 84), dtype=torch.uint8, device="cuda"),
+            "reward": torch.randn(T + 1, B, device="cuda").clamp(-1, 1),
+            "prev_action": torch.randint(0, A, (T + 1, B), device="cuda"),
+            "done": torch.rand(T + 1, B, device="cuda") > 0.95,
+        }
+        behavior_logits = torch.randn(T, B, A, device="cuda")
+        actions = torch.randint(0, A, (T, B), device="cuda")
+
+        def grads(model):
+            model.zero_grad(set_to_none=True)
+            out, _ = model(batch, tuple())
+            bootstrap = out["baseline"][-1].float()
+            logits = out["policy_logits"][:-1].float()
+            baseline = out["baseline"][:-1].float()
+            rewards = batch["reward"][1:].float()
+            discounts = (~batch["done"][1:]).float() * 0.99
+            vt = vtrace.from_logits(
+                behavior_policy_logits=behavior_logits,
+                target_policy_logits=logits.detach(),
+                actions=actions,
+                discounts=discounts,
+                rewards=rewards,
+                values=baseline.detach(),
+                bootstrap_value=bootstrap.detach(),
+            )
+            loss = fused_loss.impala_total_loss(
+                logits, baseline, actions, vt.pg_advantages, vt.vs, 0.0006, 0.5
+            )
+            loss.backward()
+            return {
+                n: p.grad.detach().float().clone()
+                for n, p in model.named_parameters()
+                if p.grad is not None
+            }
+
+        g32 = grads(m32)
+        g16 = grads(m16)
+        torch.cuda.synchronize()
+        assert set(g32) == set(g16)
+        bad = []
+        for n in g32:
+            a, b = g32[n].flatten(), g16[n].flatten()
+            denom = a.norm() * b.norm()
+            if denom < 1e-12:
+                continue
+            cos = float((a @ b) / denom)
+            if cos < 0.9:
+                bad.append((n, cos))
+        assert not bad, bad
