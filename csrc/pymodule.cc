@@ -124,40 +124,81 @@ class RpcWrapper {
     return rpc_->localAddrs();
   }
   void setTransports(py::object transports) {
-    // We always run both transports (abstract unix + tcp). Accept the
-    // reference's strings for compatibility; reject unknown ones.
+    // Reference semantics (rpc.cc:324-336): named transports stay enabled,
+    // the rest are disabled — no default listener, never dialed. Accept the
+    // reference's aliases; "shared memory"/"infiniband" map to our unix
+    // transport (memfd-backed buffers ride unix sockets; no IB verbs, same
+    // as the reference's shipped code).
+    bool tcp = false, unixSock = false;
     for (auto t : transports) {
       std::string s = py::cast<std::string>(t);
-      if (s != "tcp/ip" && s != "shared memory" && s != "uv" && s != "infiniband" && s != "tcp" &&
-          s != "unix") {
+      if (s == "tcp/ip" || s == "tcp" || s == "uv") {
+        tcp = true;
+      } else if (s == "shared memory" || s == "unix" || s == "ipc" || s == "infiniband") {
+        unixSock = true;
+      } else {
         throw RpcError("unknown transport: " + s);
       }
     }
+    rpc_->setTransports(tcp, unixSock);
+  }
+
+  // Reference ExceptionMode (src/rpc.h:201-205): what a define()d handler's
+  // exceptions do. 2 = All (default here: the error text reaches the
+  // caller), 1 = DeserializationOnly (argument-decode failures reach the
+  // caller; handler exceptions are logged and the caller times out),
+  // 0 = None (everything is only logged). The reference's C++ default is
+  // DeserializationOnly; ours is All because Python callers expect remote
+  // tracebacks — set_exception_mode restores the stricter modes.
+  void setExceptionMode(const std::string& mode) {
+    if (mode == "none") excMode_->store(0);
+    else if (mode == "deserialization_only") excMode_->store(1);
+    else if (mode == "all") excMode_->store(2);
+    else throw RpcError("exception mode must be none|deserialization_only|all");
   }
 
   void define(const std::string& name, py::function fn) {
     auto g = std::make_shared<PyGuard>(fn);
     std::weak_ptr<Rpc> wr = rpc_;
-    rpc_->define(name, [g, wr](Frame f, const std::string& from, RespondFn respond) {
+    auto mode = excMode_;
+    rpc_->define(name, [g, wr, mode](Frame f, const std::string& from, RespondFn respond) {
       if (!pyAlive()) return;
       auto rpc = wr.lock();
       bool local = rpc && rpc->peerIsLocal(from);
       std::string payload;
       std::vector<at::Tensor> tensors;
       bool isErr = false;
+      bool suppress = false;
       {
         py::gil_scoped_acquire gil;
+        int m = mode->load();
         try {
           auto [args, kwargs] = deserializeCall(f.payload, f.tensors);
-          py::object r = g->obj(*args, **kwargs);
-          payload = serializeObject(r, tensors, local);
+          try {
+            py::object r = g->obj(*args, **kwargs);
+            payload = serializeObject(r, tensors, local);
+          } catch (py::error_already_set& e) {
+            payload = e.what();
+            isErr = true;
+            suppress = m < 2;
+          } catch (const std::exception& e) {
+            payload = e.what();
+            isErr = true;
+            suppress = m < 2;
+          }
         } catch (py::error_already_set& e) {
           payload = e.what();
           isErr = true;
+          suppress = m < 1;
         } catch (const std::exception& e) {
           payload = e.what();
           isErr = true;
+          suppress = m < 1;
         }
+      }
+      if (isErr && suppress) {
+        MRL_LOG_ERROR("handler error (suppressed by exception mode): %s", payload.c_str());
+        return;  // caller times out, reference None/DeserializationOnly behavior
       }
       respond(std::move(payload), std::move(tensors), isErr);
     });
@@ -266,6 +307,7 @@ class RpcWrapper {
 
  private:
   RpcPtr rpc_;
+  std::shared_ptr<std::atomic<int>> excMode_ = std::make_shared<std::atomic<int>>(2);
 };
 
 // -------------------------------------------------------------- Broker
@@ -564,6 +606,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("get_name", &RpcWrapper::getName)
       .def("set_timeout", &RpcWrapper::setTimeout, py::arg("timeout"))
       .def("set_transports", &RpcWrapper::setTransports, py::arg("transports"))
+      .def("set_exception_mode", &RpcWrapper::setExceptionMode, py::arg("mode"))
       .def("listen", &RpcWrapper::listen, py::arg("address"))
       .def("connect", &RpcWrapper::connect, py::arg("address"))
       .def("local_addrs", &RpcWrapper::localAddrs)

@@ -85,22 +85,40 @@ std::string Rpc::getName() const {
 void Rpc::ensureListeningLocked() {
   if (defaultListenersCreated_) return;
   defaultListenersCreated_ = true;
-  try {
-    auto tcp = engine_->listen("tcp://0.0.0.0:0");
-    for (auto& a : tcp) listenAddrs_.push_back(a);
-  } catch (const std::exception& e) {
-    MRL_LOG_ERROR("default tcp listener failed: %s", e.what());
+  if (tcpEnabled_) {
+    try {
+      auto tcp = engine_->listen("tcp://0.0.0.0:0");
+      for (auto& a : tcp) listenAddrs_.push_back(a);
+    } catch (const std::exception& e) {
+      MRL_LOG_ERROR("default tcp listener failed: %s", e.what());
+    }
   }
-  try {
-    auto ux = engine_->listen("unix://" + machineId_ + "/" + uid_);
-    for (auto& a : ux) listenAddrs_.push_back(a);
-  } catch (const std::exception& e) {
-    MRL_LOG_ERROR("default unix listener failed: %s", e.what());
+  if (unixEnabled_) {
+    try {
+      auto ux = engine_->listen("unix://" + machineId_ + "/" + uid_);
+      for (auto& a : ux) listenAddrs_.push_back(a);
+    } catch (const std::exception& e) {
+      MRL_LOG_ERROR("default unix listener failed: %s", e.what());
+    }
   }
+}
+
+void Rpc::setTransports(bool tcp, bool unixSock) {
+  std::lock_guard<std::mutex> lk(mu_);
+  if (defaultListenersCreated_) {
+    throw RpcError("set_transports must be called before listen/connect");
+  }
+  if (!tcp && !unixSock) throw RpcError("at least one transport must stay enabled");
+  tcpEnabled_ = tcp;
+  unixEnabled_ = unixSock;
 }
 
 std::vector<std::string> Rpc::listen(const std::string& addr) {
   std::lock_guard<std::mutex> lk(mu_);
+  bool isUnix = addr.rfind("unix://", 0) == 0;
+  if ((isUnix && !unixEnabled_) || (!isUnix && !tcpEnabled_)) {
+    throw RpcError("transport disabled by set_transports: " + addr);
+  }
   auto bound = engine_->listen(addr);
   for (auto& a : bound) listenAddrs_.push_back(a);
   ensureListeningLocked();
@@ -325,11 +343,11 @@ void Rpc::tryConnectPeerLocked(const std::string& name, PeerInfo& p) {
   std::string addr;
   double best = 0;
   for (auto& cand : p.addrs) {
+    bool isUnix = cand.rfind("unix://", 0) == 0;
+    if ((isUnix && !unixEnabled_) || (!isUnix && !tcpEnabled_)) continue;
     TransportStat& st = p.transport[cand];
     st.failPenalty *= 0.5;  // dead transports get retried eventually
-    double score = st.samples == 0
-                       ? (cand.rfind("unix://", 0) == 0 ? -2.0 : -1.0)
-                       : st.ema;
+    double score = st.samples == 0 ? (isUnix ? -2.0 : -1.0) : st.ema;
     score += st.failPenalty;
     if (addr.empty() || score < best) {
       best = score;

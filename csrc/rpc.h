@@ -75,6 +75,11 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
   // not-yet-greeted peers report false (callers fall back to staging).
   bool peerIsLocal(const std::string& peerName);
 
+  // Restrict transports (reference rpc.cc:324-336 semantics): disabled
+  // transports get no default listener and are never dialed. Must be
+  // called before the first listen/connect.
+  void setTransports(bool tcp, bool unixSock);
+
   void shutdown();
   bool isShutdown() const { return stopping_.load(); }
 
@@ -198,6 +203,8 @@ class Rpc : public std::enable_shared_from_this<Rpc> {
   std::vector<Endpoint> endpoints_;
   std::vector<std::string> listenAddrs_;
   bool defaultListenersCreated_ = false;
+  bool tcpEnabled_ = true;
+  bool unixEnabled_ = true;
 
   std::unique_ptr<SocketEngine> engine_;
   std::thread timerThread_;

@@ -494,3 +494,88 @@ class TestInferenceServerExample:
         mean_b = sum(batch_sizes) / len(batch_sizes)
         assert mean_b > 1.3, mean_b  # singles-forever would be 1.0
         assert max(batch_sizes) >= 3
+
+
+class TestTransportRestriction:
+    def test_tcp_only(self):
+        host = moolib_amd.Rpc()
+        host.set_name("tcphost")
+        host.set_transports(["tcp/ip"])
+        addrs = host.listen("127.0.0.1:0")
+        assert not any(a.startswith("unix://") for a in host.local_addrs())
+        client = moolib_amd.Rpc()
+        client.set_name("tcpclient")
+        client.set_timeout(10)
+        host.define("f", lambda: "ok")
+        client.connect(addrs[0])
+        assert client.sync("tcphost", "f") == "ok"
+
+    def test_restrict_after_listen_rejected(self):
+        r = moolib_amd.Rpc()
+        r.set_name("late")
+        r.listen("127.0.0.1:0")
+        with pytest.raises(Exception, match="before listen"):
+            r.set_transports(["tcp/ip"])
+
+    def test_disabled_scheme_listen_rejected(self):
+        r = moolib_amd.Rpc()
+        r.set_name("noux")
+        r.set_transports(["tcp/ip"])
+        with pytest.raises(Exception, match="disabled"):
+            r.listen("unix://something")
+
+
+class TestExceptionModes:
+    def _pair(self):
+        host = moolib_amd.Rpc()
+        host.set_name("emhost")
+        addr = host.listen("127.0.0.1:0")[0]
+        client = moolib_amd.Rpc()
+        client.set_name("emclient")
+        client.set_timeout(2)
+        client.connect(addr)
+        return host, client
+
+    def test_all_forwards_handler_error(self):
+        host, client = self._pair()
+
+        def boom():
+            raise ValueError("xyzzy")
+
+        host.define("boom", boom)  # default mode: all
+        with pytest.raises(Exception, match="xyzzy"):
+            client.sync("emhost", "boom")
+
+    def test_deserialization_only_suppresses_handler_error(self):
+        host, client = self._pair()
+        host.set_exception_mode("deserialization_only")
+
+        def boom():
+            raise ValueError("xyzzy")
+
+        host.define("boom", boom)
+        with pytest.raises(Exception, match="timed out"):
+            client.sync("emhost", "boom")
+        # deserialization failures still reach the caller: an argument only
+        # unpicklable on the host side
+        host.define("takes", lambda x: "got")
+
+        class Weird:
+            def __reduce__(self):
+                return (eval, ("__import__('missing_module_xyz')",))
+
+        # the decode failure is forwarded (exact text depends on where the
+        # embedded unpickle fails), NOT suppressed like the handler error
+        with pytest.raises(Exception):
+            client.sync("emhost", "takes", Weird())
+
+    def test_mode_none_suppresses_everything(self):
+        host, client = self._pair()
+        host.set_exception_mode("none")
+
+        def boom():
+            raise ValueError("xyzzy")
+
+        host.define("boom", boom)
+        with pytest.raises(Exception, match="timed out"):
+            client.sync("emhost", "boom")
