@@ -54,7 +54,9 @@ def main():
         raise SystemExit("run under torch.distributed.run with >=2 ranks")
     dist.init_process_group(backend, rank=rank, world_size=world)
     if use_cuda and backend == "nccl":
-        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        # modulo: a 1-GPU smoke with 2 ranks shares the device (RCCL may
+        # refuse duplicate-device communicators; the smoke reports that)
+        local_rank = int(os.environ.get("LOCAL_RANK", "0")) % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
     else:

@@ -559,7 +559,7 @@ class TestBatchedCopy:
         srcs = [torch.randint(0, 255, (16, 84, 84, 4), dtype=torch.uint8, device="cuda") for _ in range(3)]
         cases += [(big.select(0, i), s) for i, s in enumerate(srcs)]
         # cat-style: narrow along dim 1 (strided rows)
-        tgt = torch.empty(21, 32, 7, device="cuda")
+        tgt = torch.empty(21, 40, 7, device="cuda")
         src = torch.randn(21, 128, 7, device="cuda")
         cases.append((tgt.narrow(1, 4, 32), src.narrow(1, 96, 32)))
         # odd rowBytes -> byte path
