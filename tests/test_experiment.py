@@ -36,3 +36,47 @@ def test_experiment_runs_and_resumes(tmp_path):
     out2 = subprocess.run(cmd2, capture_output=True, text=True, timeout=240, cwd=ROOT, env=env)
     assert out2.returncode == 0, out2.stderr[-3000:]
     assert "loaded checkpoint" in out2.stderr
+
+
+class TestSlurmLauncher:
+    def test_build_sbatch_structure(self):
+        import sys
+        sys.path.insert(0, "examples")
+        import launch_slurm
+
+        argv = launch_slurm.build_sbatch(
+            8, "10.0.0.1:4431", "/save/dir", job_name="p/g", partition="mi355x"
+        )
+        assert argv[0] == "sbatch"
+        assert argv[argv.index("--array") + 1] == "0-7"
+        assert argv[argv.index("--gpus-per-task") + 1] == "1"
+        assert argv[argv.index("--partition") + 1] == "mi355x"
+        wrap = argv[argv.index("--wrap") + 1]
+        assert "--connect 10.0.0.1:4431" in wrap
+        assert "--savedir /save/dir" in wrap
+
+    def test_broker_probe(self, tmp_path):
+        import sys
+        sys.path.insert(0, "examples")
+        import launch_slurm
+
+        import moolib_amd
+
+        broker_rpc = moolib_amd.Rpc()
+        broker_rpc.set_name("broker")
+        addr = [a for a in broker_rpc.listen("127.0.0.1:0") if a.startswith("tcp://")][0]
+        hostport = addr[len("tcp://"):]
+        assert launch_slurm.broker_is_alive(hostport, timeout=5)
+        assert not launch_slurm.broker_is_alive("127.0.0.1:1", timeout=1.5)
+
+    def test_dry_run(self, tmp_path, capsys):
+        import sys
+        sys.path.insert(0, "examples")
+        import launch_slurm
+
+        rc = launch_slurm.main(
+            ["-n", "2", "--dry", "--no-checks", "--savedir", str(tmp_path / "sv")]
+        )
+        assert rc == 0
+        out = capsys.readouterr().out
+        assert "sbatch" in out and "--array 0-1" in out
