@@ -54,9 +54,18 @@ def main():
         raise SystemExit("run under torch.distributed.run with >=2 ranks")
     dist.init_process_group(backend, rank=rank, world_size=world)
     if use_cuda and backend == "nccl":
-        # modulo: a 1-GPU smoke with 2 ranks shares the device (RCCL may
-        # refuse duplicate-device communicators; the smoke reports that)
-        local_rank = int(os.environ.get("LOCAL_RANK", "0")) % torch.cuda.device_count()
+        local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+        if local_rank >= torch.cuda.device_count():
+            # Verified on MI355X/ROCm 7: RCCL (2.26) refuses duplicate-device
+            # communicators ("Duplicate GPU detected"), so a shared-GPU RCCL
+            # smoke is impossible by design — use --backend gloo for a
+            # 1-GPU multi-rank plumbing check; real RCCL numbers need one
+            # GPU per rank.
+            raise SystemExit(
+                "rank %d has no dedicated GPU (%d visible): RCCL requires one "
+                "GPU per rank; rerun with --backend gloo for a plumbing smoke"
+                % (local_rank, torch.cuda.device_count())
+            )
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
     else:
