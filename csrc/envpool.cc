@@ -1,5 +1,6 @@
 #include "envpool.h"
 
+#include <fcntl.h>
 #include <semaphore.h>
 #include <signal.h>
 #include <sys/mman.h>
@@ -12,6 +13,7 @@
 
 #include <atomic>
 #include <cstring>
+#include <thread>
 
 #include "common.h"
 #include "message.h"
@@ -34,9 +36,18 @@ struct FieldDesc {
   uint64_t bytesPerEnv;
 };
 
+constexpr uint32_t kSegMagic = 0x6d726c45;  // 'mrlE'
+
 struct Header {
+  std::atomic<uint32_t> magicReady;  // kSegMagic once the creator's init is done
   std::atomic<uint32_t> layoutReady;
   std::atomic<uint32_t> workerFailed;  // worker index + 1
+  std::atomic<uint32_t> claimedWorkers;  // external EnvRunners claim slots
+  // Geometry, so a named-segment attacher can recompute the layout.
+  int32_t numWorkers;
+  int32_t batchSize;
+  int32_t numBatches;
+  int64_t segBytes;
   uint32_t numFields;
   uint64_t bytesPerBatch;
   FieldDesc fields[kMaxFields];
@@ -71,38 +82,46 @@ int64_t dtypeBytes(uint8_t code) {
 class EnvPoolImpl : public std::enable_shared_from_this<EnvPoolImpl> {
  public:
   EnvPoolImpl(py::object createEnv, int numProcesses, int batchSize, int numBatches,
-              int64_t segBytes)
+              int64_t segBytes, std::string shmName = "", bool externalWorkers = false)
       : createEnv_(std::move(createEnv)),
         numWorkers_(numProcesses),
         batchSize_(batchSize),
-        numBatches_(numBatches) {
+        numBatches_(numBatches),
+        shmName_(shmPath(shmName)),
+        externalWorkers_(externalWorkers) {
     if (numWorkers_ <= 0 || batchSize_ <= 0 || numBatches_ <= 0) {
       throw RpcError("envpool: sizes must be positive");
     }
+    if (externalWorkers_ && shmName_.empty()) {
+      throw RpcError("envpool: external workers need a shm_name to attach to");
+    }
     if (numWorkers_ > batchSize_) numWorkers_ = batchSize_;
 
-    // ---- carve the fixed regions ----
-    size_t off = 0;
-    auto alloc = [&](size_t n, size_t align = 64) {
-      off = (off + align - 1) & ~(align - 1);
-      size_t at = off;
-      off += n;
-      return at;
-    };
-    headerOff_ = alloc(sizeof(Header));
-    actionsOff_ = alloc(sizeof(int64_t) * numBatches_ * batchSize_);
-    queuesOff_ = alloc(sizeof(WorkerQueue) * numWorkers_);
-    doneSemsOff_ = alloc(sizeof(sem_t) * numBatches_ * numWorkers_);
-    dataOff_ = alloc(1, 4096);
+    computeLayout();
     if (segBytes < static_cast<int64_t>(dataOff_) + (64 << 20)) segBytes = dataOff_ + (512 << 20);
     segBytes_ = segBytes;
 
-    base_ = static_cast<char*>(
-        mmap(nullptr, segBytes_, PROT_READ | PROT_WRITE, MAP_SHARED | MAP_ANONYMOUS, -1, 0));
-    if (base_ == MAP_FAILED) throw RpcError("envpool: mmap failed");
+    if (shmName_.empty()) {
+      base_ = static_cast<char*>(
+          mmap(nullptr, segBytes_, PROT_READ | PROT_WRITE, MAP_SHARED | MAP_ANONYMOUS, -1, 0));
+      if (base_ == MAP_FAILED) throw RpcError("envpool: mmap failed");
+    } else {
+      shmFd_ = shm_open(shmName_.c_str(), O_CREAT | O_EXCL | O_RDWR, 0600);
+      if (shmFd_ < 0) throw RpcError("envpool: shm_open(create " + shmName_ + ") failed");
+      shmOwner_ = true;
+      if (ftruncate(shmFd_, segBytes_) != 0) throw RpcError("envpool: ftruncate failed");
+      base_ = static_cast<char*>(
+          mmap(nullptr, segBytes_, PROT_READ | PROT_WRITE, MAP_SHARED, shmFd_, 0));
+      if (base_ == MAP_FAILED) throw RpcError("envpool: mmap(shm) failed");
+    }
     std::memset(base_, 0, dataOff_);
 
     new (header()) Header();
+    Header* h = header();
+    h->numWorkers = numWorkers_;
+    h->batchSize = batchSize_;
+    h->numBatches = numBatches_;
+    h->segBytes = segBytes_;
     for (int w = 0; w < numWorkers_; ++w) {
       WorkerQueue* q = queue(w);
       sem_init(&q->itemsSem, 1, 0);
@@ -110,6 +129,9 @@ class EnvPoolImpl : public std::enable_shared_from_this<EnvPoolImpl> {
       q->tail.store(0);
     }
     for (int i = 0; i < numBatches_ * numWorkers_; ++i) sem_init(&doneSems()[i], 1, 0);
+    h->magicReady.store(kSegMagic, std::memory_order_release);
+
+    if (externalWorkers_) return;  // EnvRunner processes will claim the slots
 
     // ---- fork the workers (GIL is held: safe point for PyOS_AfterFork) ----
     for (int w = 0; w < numWorkers_; ++w) {
@@ -132,16 +154,115 @@ class EnvPoolImpl : public std::enable_shared_from_this<EnvPoolImpl> {
     }
   }
 
+  // Attach to a named segment created by another process (EnvRunner side).
+  struct AttachTag {};
+  EnvPoolImpl(AttachTag, py::object createEnv, const std::string& shmName)
+      : createEnv_(std::move(createEnv)), shmName_(shmPath(shmName)), attached_(true) {
+    TimePoint t0 = now();
+    while (true) {
+      shmFd_ = shm_open(shmName_.c_str(), O_RDWR, 0600);
+      if (shmFd_ >= 0) break;
+      if (secondsSince(t0) > 60.0) {
+        throw RpcError("envrunner: no segment named " + shmName_ + " appeared within 60s");
+      }
+      usleep(20000);
+    }
+    // Map the header, wait for init, read the geometry, then map fully.
+    void* peek = mmap(nullptr, sizeof(Header), PROT_READ, MAP_SHARED, shmFd_, 0);
+    if (peek == MAP_FAILED) throw RpcError("envrunner: header mmap failed");
+    Header* ph = reinterpret_cast<Header*>(peek);
+    while (ph->magicReady.load(std::memory_order_acquire) != kSegMagic) {
+      if (secondsSince(t0) > 60.0) throw RpcError("envrunner: segment never initialized");
+      usleep(1000);
+    }
+    numWorkers_ = ph->numWorkers;
+    batchSize_ = ph->batchSize;
+    numBatches_ = ph->numBatches;
+    segBytes_ = ph->segBytes;
+    munmap(peek, sizeof(Header));
+    computeLayout();
+    base_ = static_cast<char*>(
+        mmap(nullptr, segBytes_, PROT_READ | PROT_WRITE, MAP_SHARED, shmFd_, 0));
+    if (base_ == MAP_FAILED) throw RpcError("envrunner: mmap failed");
+  }
+
+  // Claim a worker slot and serve it on a host thread until terminated.
+  void startRunnerThread() {
+    int w = static_cast<int>(header()->claimedWorkers.fetch_add(1));
+    if (w >= numWorkers_) {
+      header()->claimedWorkers.fetch_sub(1);
+      throw RpcError("envrunner: all " + std::to_string(numWorkers_) + " slots claimed");
+    }
+    runnerThread_ = std::thread([this, w] {
+      py::gil_scoped_acquire gil;
+      workerMain(w);
+    });
+  }
+
   ~EnvPoolImpl() {
+    terminate_.store(true);
+    if (runnerThread_.joinable()) {
+      // The runner thread re-acquires the GIL between queue waits; joining
+      // while holding it would deadlock.
+      if (PyGILState_Check()) {
+        py::gil_scoped_release rel;
+        runnerThread_.join();
+      } else {
+        runnerThread_.join();
+      }
+    }
     for (pid_t p : pids_) kill(p, SIGKILL);
     for (pid_t p : pids_) waitpid(p, nullptr, 0);
     if (base_ && base_ != MAP_FAILED) munmap(base_, segBytes_);
+    if (shmFd_ >= 0) close(shmFd_);
+    if (shmOwner_) shm_unlink(shmName_.c_str());
   }
+
+  bool runnerAlive() { return runnerThread_.joinable() && !terminate_.load(); }
+
+ private:
+  static std::string shmPath(const std::string& name) {
+    return name.empty() ? name : "/mrl-env-" + name;
+  }
+
+  void computeLayout() {
+    size_t off = 0;
+    auto alloc = [&](size_t n, size_t align = 64) {
+      off = (off + align - 1) & ~(align - 1);
+      size_t at = off;
+      off += n;
+      return at;
+    };
+    headerOff_ = alloc(sizeof(Header));
+    actionsOff_ = alloc(sizeof(int64_t) * numBatches_ * batchSize_);
+    queuesOff_ = alloc(sizeof(WorkerQueue) * numWorkers_);
+    doneSemsOff_ = alloc(sizeof(sem_t) * numBatches_ * numWorkers_);
+    dataOff_ = alloc(1, 4096);
+  }
+
+ public:
 
   // ------------------------------------------------------------- client
 
   void step(int b, py::object action) {
     if (b < 0 || b >= numBatches_) throw RpcError("envpool: bad batch index");
+    if (externalWorkers_ && !allClaimed_) {
+      // Work posted before every slot has a runner would wait forever on
+      // the unclaimed slots' completion semaphores.
+      py::gil_scoped_release rel;
+      TimePoint t0 = now();
+      while (header()->claimedWorkers.load(std::memory_order_acquire) <
+             static_cast<uint32_t>(numWorkers_)) {
+        checkFailureNoGil();
+        if (secondsSince(t0) > 120.0) {
+          throw RpcError("envpool: only " +
+                         std::to_string(header()->claimedWorkers.load()) + "/" +
+                         std::to_string(numWorkers_) + " EnvRunner slots claimed after 120s");
+        }
+        usleep(5000);
+      }
+      allClaimed_ = true;
+    }
     at::Tensor a = THPVariable_Unpack(action.ptr());
     if (!a.device().is_cpu()) a = a.to(at::kCPU);
     a = a.to(at::kLong).contiguous();
@@ -184,12 +305,14 @@ class EnvPoolImpl : public std::enable_shared_from_this<EnvPoolImpl> {
   }
 
   bool anyAlive() {
+    if (externalWorkers_) return header()->claimedWorkers.load() > 0;
     for (pid_t p : pids_) {
       if (waitpid(p, nullptr, WNOHANG) == 0) return true;
     }
     return false;
   }
   int aliveCount() {
+    if (externalWorkers_) return static_cast<int>(header()->claimedWorkers.load());
     int n = 0;
     for (pid_t p : pids_) {
       if (waitpid(p, nullptr, WNOHANG) == 0) ++n;
@@ -392,8 +515,22 @@ class EnvPoolImpl : public std::enable_shared_from_this<EnvPoolImpl> {
       while (true) {
         {
           py::gil_scoped_release rel;
-          while (sem_wait(&q->itemsSem) != 0) {
-            if (errno != EINTR) throw RpcError("envpool: sem_wait failed");
+          bool got = false;
+          while (!got) {
+            timespec ts;
+            clock_gettime(CLOCK_REALTIME, &ts);
+            ts.tv_nsec += 250 * 1000 * 1000;
+            if (ts.tv_nsec >= 1000000000) {
+              ts.tv_sec += 1;
+              ts.tv_nsec -= 1000000000;
+            }
+            if (sem_timedwait(&q->itemsSem, &ts) == 0) {
+              got = true;
+            } else if (errno == ETIMEDOUT) {
+              if (terminate_.load(std::memory_order_relaxed)) return;
+            } else if (errno != EINTR) {
+              throw RpcError("envpool: sem_wait failed");
+            }
           }
         }
         uint32_t head = q->head.load(std::memory_order_relaxed);
@@ -412,15 +549,24 @@ class EnvPoolImpl : public std::enable_shared_from_this<EnvPoolImpl> {
   }
 
   py::object createEnv_;
-  int numWorkers_;
-  int batchSize_;
-  int numBatches_;
+  int numWorkers_ = 0;
+  int batchSize_ = 0;
+  int numBatches_ = 0;
   int64_t segBytes_ = 0;
   char* base_ = nullptr;
   size_t headerOff_ = 0, actionsOff_ = 0, queuesOff_ = 0, doneSemsOff_ = 0, dataOff_ = 0;
   std::vector<pid_t> pids_;
   bool tensorsBuilt_ = false;
   std::vector<py::object> batchDicts_;
+  // named-segment / external-worker mode
+  std::string shmName_;
+  bool externalWorkers_ = false;
+  bool attached_ = false;
+  bool allClaimed_ = false;
+  bool shmOwner_ = false;
+  int shmFd_ = -1;
+  std::atomic<bool> terminate_{false};
+  std::thread runnerThread_;
 };
 
 // ------------------------------------------------------------- wrappers
@@ -431,12 +577,26 @@ py::object EnvStepperFuture::result() {
 }
 
 EnvPool::EnvPool(py::object createEnv, int numProcesses, int batchSize, int numBatches,
-                 int64_t sharedMemoryBytes) {
+                 int64_t sharedMemoryBytes, const std::string& shmName, bool externalWorkers) {
   impl_ = std::make_shared<EnvPoolImpl>(std::move(createEnv), numProcesses, batchSize, numBatches,
-                                        sharedMemoryBytes);
+                                        sharedMemoryBytes, shmName, externalWorkers);
 }
 
 EnvPool::~EnvPool() = default;
+
+EnvRunner::EnvRunner(py::object createEnv) : createEnv_(std::move(createEnv)) {}
+
+EnvRunner::~EnvRunner() = default;
+
+void EnvRunner::start(const std::string& shmName) {
+  if (impl_) throw RpcError("envrunner: already started");
+  // GIL stays held: the segment creator is another process, so the wait
+  // does not depend on our Python threads.
+  impl_ = std::make_shared<EnvPoolImpl>(EnvPoolImpl::AttachTag{}, createEnv_, shmName);
+  impl_->startRunnerThread();
+}
+
+bool EnvRunner::running() { return impl_ && impl_->runnerAlive(); }
 
 EnvStepperFuture EnvPool::step(int batchIndex, py::object action) {
   impl_->step(batchIndex, action);

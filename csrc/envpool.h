@@ -42,8 +42,12 @@ class EnvStepperFuture {
 
 class EnvPool {
  public:
+  // shmName nonempty: the segment is shm_open()-named so OTHER processes
+  // can host the workers (see EnvRunner); externalWorkers then skips the
+  // fork and waits for runners to claim the num_processes slots.
   EnvPool(py::object createEnv, int numProcesses, int batchSize, int numBatches,
-          int64_t sharedMemoryBytes);
+          int64_t sharedMemoryBytes, const std::string& shmName = "",
+          bool externalWorkers = false);
   ~EnvPool();
   EnvStepperFuture step(int batchIndex, py::object action);
   at::Tensor sharedBuffer();
@@ -51,6 +55,23 @@ class EnvPool {
   int numWorkersAlive();
 
  private:
+  std::shared_ptr<EnvPoolImpl> impl_;
+};
+
+// Hosts env-stepping capacity for an EnvPool created with shm_name in
+// ANOTHER process on this machine — the reference's EnvRunner role
+// (src/env.h:363-453: attach to a named segment, claim a client slot, run
+// the worker loop). Ours runs the loop on a thread of the calling process
+// (so the runner can be any separately-launched python program).
+class EnvRunner {
+ public:
+  explicit EnvRunner(py::object createEnv);
+  ~EnvRunner();
+  void start(const std::string& shmName);
+  bool running();
+
+ private:
+  py::object createEnv_;
   std::shared_ptr<EnvPoolImpl> impl_;
 };
 

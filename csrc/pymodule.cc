@@ -642,13 +642,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("result", &EnvStepperFuture::result);
 
   py::class_<EnvPool>(m, "EnvPool")
-      .def(py::init<py::object, int, int, int, int64_t>(), py::arg("create_env"),
-           py::arg("num_processes"), py::arg("batch_size"), py::arg("num_batches"),
-           py::arg("shared_memory_bytes") = 0)
+      .def(py::init<py::object, int, int, int, int64_t, const std::string&, bool>(),
+           py::arg("create_env"), py::arg("num_processes"), py::arg("batch_size"),
+           py::arg("num_batches"), py::arg("shared_memory_bytes") = 0,
+           py::arg("shm_name") = std::string(), py::arg("external_workers") = false)
       .def("step", &EnvPool::step, py::arg("batch_index"), py::arg("action"))
       .def("shared_buffer", &EnvPool::sharedBuffer)
       .def("running", &EnvPool::running)
       .def("num_workers_alive", &EnvPool::numWorkersAlive);
+
+  py::class_<EnvRunner>(m, "EnvRunner")
+      .def(py::init<py::object>(), py::arg("create_env"))
+      .def("start", &EnvRunner::start, py::arg("shm_name"),
+           "attach to the named EnvPool segment and serve one worker slot")
+      .def("running", &EnvRunner::running);
 
   py::class_<Batcher>(m, "Batcher")
       .def(py::init<int64_t, py::object, int64_t>(), py::arg("size"),

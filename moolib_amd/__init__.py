@@ -401,22 +401,10 @@ class EnvStepper:
         return self._pool.step(batch_index, action)
 
 
-class EnvRunner:
-    """API-parity shim for the reference's EnvRunner (src/env.h:363-405).
-
-    The reference uses EnvRunner to host env workers reached over RPC; in
-    this implementation workers are forked by EnvPool directly, so the
-    runner only reports liveness.
-    """
-
-    def __init__(self, pool=None):
-        self._pool = pool
-
-    def start(self):
-        return None
-
-    def running(self):
-        return self._pool.running() if self._pool is not None else False
+# Real env-server role (reference src/env.h:363-453): EnvRunner(create_env)
+# .start(shm_name) attaches to an EnvPool created elsewhere with
+# shm_name=... + external_workers=True and serves one worker slot.
+from ._core import EnvRunner  # noqa: F401
 
 
 atexit.register(_core._shutdown_all)

@@ -192,3 +192,81 @@ class TestMidEpisodeError:
         with _pt.raises(RuntimeError, match="exploded|died|failed"):
             for i in range(10):
                 pool.step(0, torch.zeros(2, dtype=torch.int64)).result()
+
+
+def _runner_proc(shm_name, stop_file):
+    import os
+    import time as _t
+
+    import moolib_amd as M
+
+    runner = M.EnvRunner(CountingEnv)
+    runner.start(shm_name)
+    t0 = _t.time()
+    while not os.path.exists(stop_file) and _t.time() - t0 < 120:
+        _t.sleep(0.05)
+
+
+class TestExternalEnvRunner:
+    def test_runners_in_separate_processes(self, tmp_path):
+        """EnvPool(shm_name=..., external_workers=True) is served by
+        EnvRunner processes launched independently (reference env-server
+        role, src/env.h:363-453)."""
+        import multiprocessing as mp
+        import uuid
+
+        name = "t" + uuid.uuid4().hex[:10]
+        stop_file = str(tmp_path / "stop")
+        ctx = mp.get_context("spawn")
+        procs = [
+            ctx.Process(target=_runner_proc, args=(name, stop_file)) for _ in range(2)
+        ]
+        for p in procs:
+            p.start()
+        try:
+            pool = moolib_amd.EnvPool(
+                CountingEnv,
+                num_processes=2,
+                batch_size=4,
+                num_batches=2,
+                shm_name=name,
+                external_workers=True,
+            )
+            actions = torch.arange(4, dtype=torch.int64)
+            obs = pool.step(0, torch.zeros(4, dtype=torch.int64)).result()
+            assert torch.equal(obs["state"][:, 0], torch.zeros(4))
+            for i in range(1, 5):
+                obs = pool.step(0, actions).result()
+                assert torch.equal(obs["state"][:, 0], torch.full((4,), float(i)))
+                assert torch.equal(obs["reward"], actions.float())
+            obs = pool.step(0, actions).result()  # 5th env step ends the episode
+            assert obs["done"].all()
+            # the second buffer works too
+            obs1 = pool.step(1, torch.zeros(4, dtype=torch.int64)).result()
+            assert torch.equal(obs1["state"][:, 0], torch.zeros(4))
+            assert pool.num_workers_alive() == 2
+            del pool
+        finally:
+            open(stop_file, "w").write("x")
+            for p in procs:
+                p.join(timeout=30)
+                if p.is_alive():
+                    p.kill()
+
+    def test_slot_overclaim_rejected(self, tmp_path):
+        import uuid
+
+        name = "t" + uuid.uuid4().hex[:10]
+        pool = moolib_amd.EnvPool(
+            CountingEnv, num_processes=1, batch_size=2, num_batches=1,
+            shm_name=name, external_workers=True,
+        )
+        r1 = moolib_amd.EnvRunner(CountingEnv)
+        r1.start(name)
+        r2 = moolib_amd.EnvRunner(CountingEnv)
+        with pytest.raises(Exception, match="slots claimed"):
+            r2.start(name)
+        obs = pool.step(0, torch.zeros(2, dtype=torch.int64)).result()
+        assert obs["state"].shape == (2, 2)
+        assert r1.running()
+        del pool
