@@ -862,7 +862,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA 3x3/s1/p1 NHWC conv with fused relu/bias prologue + bias/relu/residual epilogue",
         py::arg("x"), py::arg("w_packed"), py::arg("k"), py::arg("relu_in") = false,
         py::arg("bias_in") = py::none(), py::arg("epi") = 0, py::arg("bias1") = py::none(),
-        py::arg("res") = py::none(), py::arg("bias2") = py::none());
+        py::arg("res") = py::none(), py::arg("bias2") = py::none(), py::arg("rt") = 0);
   m.def("lstm_fused_fwd", &lstm_fused_fwd, "fused masked LSTM sequence scan fwd (MFMA, gfx950)");
   m.def("lstm_fused_bwd", &lstm_fused_bwd, "fused masked LSTM sequence scan bwd (MFMA, gfx950)");
   m.def("vtrace_from_log_rhos", &vtrace_from_log_rhos, "fused V-trace scan (gfx950)");

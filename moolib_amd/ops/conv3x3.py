@@ -95,9 +95,10 @@ def repack(module):
 
 
 def conv3x3(x, w_packed, k, relu_in=False, bias_in=None, epi=EPI_NONE,
-            bias1=None, res=None, bias2=None):
+            bias1=None, res=None, bias2=None, rt=0):
     """out = conv3x3_s1_p1( relu(x + bias_in) if relu_in else x ) then
-    epilogue: none / +bias1 / relu(+bias1) / +bias1+res(+bias2)."""
+    epilogue: none / +bias1 / relu(+bias1) / +bias1+res(+bias2).
+    rt>0 overrides the row-tiles-per-wave pick (microbenchmarking)."""
     to_bf = lambda t: None if t is None else t.to(torch.bfloat16)
     return _k().conv3x3_nhwc_fused(
         x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last),
@@ -109,4 +110,5 @@ def conv3x3(x, w_packed, k, relu_in=False, bias_in=None, epi=EPI_NONE,
         to_bf(bias1),
         None if res is None else to_bf(res).contiguous(memory_format=torch.channels_last),
         to_bf(bias2),
+        rt,
     )

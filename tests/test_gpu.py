@@ -486,6 +486,26 @@ class TestConv3x3Fused:
         scale = ref.abs().max().item()
         assert err / scale < 0.02, (err, scale)
 
+    @pytest.mark.parametrize("rt", [1, 2, 4])
+    def test_row_tile_variants_match(self, rt):
+        """Every RT instantiation (v2 row-tiled chains) must agree with the
+        fp32 reference, including the M-tail where pixBase+64*RT > M."""
+        from moolib_amd.ops import conv3x3 as c3
+
+        torch.manual_seed(40 + rt)
+        C = K = 32
+        # H*W*N chosen so M = 3*21*21 = 1323 is NOT a multiple of 64*rt
+        x = torch.randn(3, C, 21, 21, device="cuda", dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last
+        )
+        w = torch.randn(K, C, 3, 3, device="cuda") * 0.2
+        b1 = torch.randn(K, device="cuda") * 0.5
+        y = c3.conv3x3(x, c3.pack_weight(w), K, relu_in=True, epi=c3.EPI_BIAS, bias1=b1, rt=rt)
+        ref = self._ref(x, w, relu_in=True, epi=1, bias1=b1)
+        err = (y.float() - ref).abs().max().item()
+        scale = max(ref.abs().max().item(), 1.0)
+        assert err / scale < 0.02, (rt, err, scale)
+
     def test_fused_block_variants(self):
         from moolib_amd.ops import conv3x3 as c3
 
