@@ -62,6 +62,23 @@ py::object& ipcShare() {
   return *f;
 }
 
+// moolib_amd.shm.share_if_memfd — returns a segment-identity wrapper for
+// CPU tensors living in memfd segments (the fd+offset cross-process
+// buffer identity of the reference's memfd allocator), None for ordinary
+// tensors. Same kill switch as the hipIpc path.
+py::object& memfdShare() {
+  static py::object* f = []() -> py::object* {
+    if (std::getenv("MOOLIB_AMD_NO_IPC_RPC")) return new py::object();
+    try {
+      return new py::object(py::module_::import("moolib_amd.shm").attr("share_if_memfd"));
+    } catch (...) {
+      PyErr_Clear();
+      return new py::object();
+    }
+  }();
+  return *f;
+}
+
 }  // namespace
 
 void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors,
@@ -110,6 +127,23 @@ void serializePy(py::handle obj, WireWriter& w, std::vector<at::Tensor>& tensors
         return;
       } catch (...) {
         PyErr_Clear();  // e.g. unshareable storage — fall back to staging
+      }
+    }
+    if (t.device().is_cpu() && ipcLocal && memfdShare()) {
+      // memfd-backed tensor to a same-machine peer: ship the segment
+      // identity (pid, fd, offset); the receiver re-opens /proc/<pid>/fd
+      // and maps the same pages (reference memfd.cc capability).
+      try {
+        py::object shared =
+            memfdShare()(py::reinterpret_borrow<py::object>(obj));
+        if (!shared.is_none()) {
+          w.u8(tPickle);
+          py::bytes b = pickleDumps()(shared, 2);
+          w.str(std::string_view(PyBytes_AS_STRING(b.ptr()), PyBytes_GET_SIZE(b.ptr())));
+          return;
+        }
+      } catch (...) {
+        PyErr_Clear();  // fall through to byte transfer
       }
     }
     if (t.device().is_cpu()) {
