@@ -6,7 +6,11 @@ F.conv2d (MIOpen, after its find settles), with relu fused in ours and
 counted for MIOpen. Run on an MI355X:
     python tools/conv3x3_micro.py > profiles/rXX_conv3x3_micro.txt
 """
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
