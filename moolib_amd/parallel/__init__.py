@@ -13,22 +13,45 @@ module is the re-designed data plane.
 import torch
 
 
-def reduce_hook(process_group=None):
+# xGMI sizing: 7 point-to-point links x ~153 GB/s per MI355X GPU; a ring
+# allreduce is per-link bound, so chunks below ~8 MiB are launch-latency
+# dominated. The IMPALA model's whole bucket is ~4 MiB (1.1 M params), so
+# ONE collective is optimal there; the chunking below exists for large
+# models where pipelining chunks lets RCCL start reducing while later
+# chunks are still being launched. (Overlapping the reduction with
+# BACKWARD itself is out of protocol scope: the Accumulator contract —
+# ours and the reference's, src/accumulator.cc:880-1003 — is
+# loss.backward() THEN reduce_gradients(), so gradients are complete
+# before the hook sees the bucket.)
+_XGMI_CHUNK_BYTES = 64 << 20
+
+
+def reduce_hook(process_group=None, chunk_bytes=_XGMI_CHUNK_BYTES):
     """Build a local-reduce hook backed by torch.distributed.all_reduce.
 
     The hook is called by the C++ Accumulator with the flat on-device
-    gradient bucket; it starts an async SUM allreduce and returns a poll
+    gradient bucket; it starts async SUM allreduce(s) and returns a poll
     function the accumulator calls from update() until the collective
-    completes.
+    completes. Buckets beyond chunk_bytes are split so RCCL pipelines the
+    chunks over xGMI.
     """
     import torch.distributed as dist
 
     def hook(flat):
-        work = dist.all_reduce(flat, op=dist.ReduceOp.SUM, group=process_group, async_op=True)
+        chunk_elems = max(1, chunk_bytes // max(flat.element_size(), 1))
+        if flat.numel() <= chunk_elems:
+            parts = [flat]
+        else:
+            parts = list(torch.split(flat, chunk_elems))
+        works = [
+            dist.all_reduce(p, op=dist.ReduceOp.SUM, group=process_group, async_op=True)
+            for p in parts
+        ]
 
         def poll():
-            if work.is_completed():
-                work.wait()  # records stream dependency for the caller
+            if all(w.is_completed() for w in works):
+                for w in works:
+                    w.wait()  # records stream dependency for the caller
                 return True
             return False
 
@@ -37,19 +60,7 @@ def reduce_hook(process_group=None):
     return hook
 
 
-def install_collective_backend(accumulator, process_group=None):
+def install_collective_backend(accumulator, process_group=None, chunk_bytes=_XGMI_CHUNK_BYTES):
     """Route the accumulator's gradient allreduce through RCCL/xGMI (or any
     initialized torch.distributed backend)."""
-    accumulator.set_local_reduce_hook(reduce_hook(process_group))
-
-
-def bucket_size_bytes_for_xgmi(world_size, link_gbps=153.0, links=7):
-    """Advisory bucket sizing for ring collectives over point-to-point xGMI.
-
-    MI355X exposes 7 xGMI links per GPU at ~153 GB/s each; a ring allreduce
-    is per-link bound, so buckets should be large enough that per-step launch
-    latency amortizes: ~8 MiB per active ring is a good floor. With a single
-    flat bucket (our default) this is moot — the whole model reduces in one
-    collective — but sharded/overlapped setups can use this.
-    """
-    return max(8 << 20, int((link_gbps * 1e9 / 153.0) * 0.0005))
+    accumulator.set_local_reduce_hook(reduce_hook(process_group, chunk_bytes))

@@ -226,7 +226,9 @@ def _dist_worker(rank, world, port, results_dir, parallel=1, rounds=1):
         acc.set_parallel_gradients(parallel)
         acc.set_virtual_batch_size(1)
     acc.connect("127.0.0.1:%d" % (port + 1))
-    parallel_mod.install_collective_backend(acc)
+    # Tiny chunk size forces the multi-chunk allreduce path (the model's
+    # flat bucket is ~68 floats; 64 bytes -> several chunks).
+    parallel_mod.install_collective_backend(acc, chunk_bytes=64)
 
     t0 = time.time()
     applied = 0
