@@ -92,6 +92,75 @@ def repack(module):
             if cache is not None:
                 cache[1].copy_(pack_weight(m.weight.detach()))
                 cache[0] = m.weight._version
+            dcache = getattr(m, "_c3_dgrad_cache", None)
+            if dcache is not None:
+                dcache[1].copy_(pack_weight_dgrad(m.weight.detach()))
+                dcache[0] = m.weight._version
+
+
+def pack_weight_dgrad(w):
+    """Packed fragments for the INPUT-gradient conv.
+
+    For stride-1 pad-1 3x3, dx = conv3x3_s1_p1(dy, W') with
+    W'[c,k,kh,kw] = W[k,c,2-kh,2-kw] (transpose + 180-degree tap rotation),
+    so the backward data pass reuses the forward kernel unchanged."""
+    wd = w.flip(2, 3).permute(1, 0, 2, 3).contiguous()
+    return pack_weight(wd)
+
+
+def dgrad_buffer(conv):
+    """Version-checked persistent dgrad-packed buffer (see packed_buffer)."""
+    w = conv.weight
+    ver = w._version
+    cache = getattr(conv, "_c3_dgrad_cache", None)
+    if cache is None:
+        conv._c3_dgrad_cache = cache = [ver, pack_weight_dgrad(w.detach())]
+        return cache[1]
+    if cache[0] != ver:
+        cache[1].copy_(pack_weight_dgrad(w.detach()))
+        cache[0] = ver
+    return cache[1]
+
+
+class _Conv3x3Fn(torch.autograd.Function):
+    """3x3/s1/p1 NHWC conv with fwd AND dgrad on the MFMA kernel.
+
+    Weight gradient goes through aten.convolution_backward (MIOpen wrw
+    igemm) — the one learner conv pass without a hand-written kernel yet.
+    Drop-in for F.conv2d(x, w, None, padding=1) on supported shapes."""
+
+    @staticmethod
+    def forward(ctx, x, weight, conv_module):
+        xb = x.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+        y = conv3x3(xb, packed_buffer(conv_module), weight.shape[0])
+        ctx.save_for_backward(xb, weight)
+        ctx.conv_module = conv_module
+        ctx.x_dtype = x.dtype
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        m = ctx.conv_module
+        dy = dy.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = conv3x3(dy, dgrad_buffer(m), w.shape[1])
+            if dx.dtype != ctx.x_dtype:
+                dx = dx.to(ctx.x_dtype)
+        if ctx.needs_input_grad[1]:
+            dw = torch.ops.aten.convolution_backward(
+                dy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                [False, True, False],
+            )[1]
+            if dw.dtype != w.dtype:  # fp32 master weights under autocast
+                dw = dw.to(w.dtype)
+        return dx, dw, None
+
+
+def conv3x3_autograd(x, conv_module):
+    """Autograd-capable conv through the MFMA kernel (learner path)."""
+    return _Conv3x3Fn.apply(x, conv_module.weight, conv_module)
 
 
 def conv3x3(x, w_packed, k, relu_in=False, bias_in=None, epi=EPI_NONE,
