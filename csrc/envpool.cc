@@ -277,6 +277,17 @@ class EnvPoolImpl : public std::enable_shared_from_this<EnvPoolImpl> {
     }
   }
 
+  // Non-consuming completion check: true iff every worker has posted batch
+  // b's completion semaphore (result(b) would return without blocking).
+  bool poll(int b) {
+    if (b < 0 || b >= numBatches_) return false;
+    for (int w = 0; w < numWorkers_; ++w) {
+      int v = 0;
+      if (sem_getvalue(&doneSems()[b * numWorkers_ + w], &v) != 0 || v < 1) return false;
+    }
+    return true;
+  }
+
   py::object result(int b) {
     {
       py::gil_scoped_release rel;
@@ -602,6 +613,8 @@ EnvStepperFuture EnvPool::step(int batchIndex, py::object action) {
   impl_->step(batchIndex, action);
   return EnvStepperFuture(impl_, batchIndex);
 }
+
+bool EnvPool::poll(int batchIndex) { return impl_->poll(batchIndex); }
 
 at::Tensor EnvPool::sharedBuffer() { return impl_->dataRegion(); }
 bool EnvPool::running() { return impl_->anyAlive(); }

@@ -50,6 +50,7 @@ class EnvPool {
           bool externalWorkers = false);
   ~EnvPool();
   EnvStepperFuture step(int batchIndex, py::object action);
+  bool poll(int batchIndex);  // true iff result(batchIndex) would not block
   at::Tensor sharedBuffer();
   bool running();
   int numWorkersAlive();

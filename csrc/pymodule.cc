@@ -647,6 +647,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
            py::arg("num_batches"), py::arg("shared_memory_bytes") = 0,
            py::arg("shm_name") = std::string(), py::arg("external_workers") = false)
       .def("step", &EnvPool::step, py::arg("batch_index"), py::arg("action"))
+      .def("poll", &EnvPool::poll, py::arg("batch_index"),
+           "true iff result(batch_index) would return without blocking")
       .def("shared_buffer", &EnvPool::sharedBuffer)
       .def("running", &EnvPool::running)
       .def("num_workers_alive", &EnvPool::numWorkersAlive);

@@ -88,6 +88,10 @@ class ImpalaConfig:
     #   with per-capture RNG generators registered; single-stream replay is
     #   9/9 stable and costs only ~2%. Re-enable with
     #   MOOLIB_AMD_FORCE_SIDE_STREAM=1 to investigate (round 2).
+    prefetch_h2d: bool = True        # DMA env results to HBM on a copy stream as soon as
+    #   the workers finish (poll + persistent device buffers + events), so the
+    #   ~2 ms/step of frame H2D overlaps learner/actor compute instead of
+    #   serializing in the act loop.
     max_learn_backlog: int = 4       # stop acting when this many learn batches are queued
     group_timeout: float = 10.0      # broker eviction / allreduce timeout (seconds)
     bf16_weights: bool = True        # forward on bf16 shadow weights (fp32 master for optimizer/sync):
@@ -111,6 +115,12 @@ class EnvBatchState:
         self.step_count = torch.zeros(B)
         self.time_batcher = moolib_amd.Batcher(cfg.unroll_length + 1, device)
         self.pinned = {}  # field -> pinned staging tensor
+        # H2D prefetch (copy-stream pipeline): persistent device buffers the
+        # prefetcher DMAs shm results into while the GPU is busy elsewhere.
+        self.staged = None           # field -> device tensor (stable addresses)
+        self.prefetched = None       # cpu result dict already consumed from the future
+        self.h2d_event = None        # recorded on the copy stream after staging
+        self.consumed_event = None   # recorded on the compute stream after last read
 
     def update(self, env_outputs, action, stats):
         # prev_action is a persistent device buffer (stable address for
@@ -305,6 +315,15 @@ class ImpalaPeer:
             if (self.is_cuda and cfg.graph_learner)
             else self._learn_fn
         )
+        self._h2d_stream = (
+            torch.cuda.Stream()
+            if (
+                self.is_cuda
+                and cfg.prefetch_h2d
+                and not os.environ.get("MOOLIB_AMD_NO_PREFETCH")
+            )
+            else None
+        )
         self._shm_registered = False
         if self.is_cuda and cfg.shm_host_register:
             try:
@@ -491,6 +510,10 @@ class ImpalaPeer:
             time.sleep(0.05)
             return "idle"
 
+        # Opportunistic H2D: if the next batch's envs are done, start the
+        # copy-stream DMA before committing this iteration to learn or act.
+        self._try_prefetch()
+
         t0 = time.perf_counter() if self.profile else 0.0
         if acc.has_gradients():
             gstats = acc.get_gradient_stats()
@@ -534,15 +557,51 @@ class ImpalaPeer:
             self.act_once()
             return "act"
 
+    def _try_prefetch(self):
+        """If the next batch's env step has completed, DMA its results into
+        persistent device buffers on the copy stream NOW — the transfer
+        overlaps whatever the compute stream is doing (learner fwd+bwd,
+        the other batch's actor forward)."""
+        if self._h2d_stream is None:
+            return
+        cur = self.next_env_index
+        st = self.env_states[cur]
+        if st.future is None or st.prefetched is not None:
+            return
+        if not self.envs.poll(cur):
+            return
+        cpu_out = st.future.result()  # poll() said this returns immediately
+        st.future = None
+        if st.h2d_event is None:
+            st.h2d_event = torch.cuda.Event()
+            st.consumed_event = torch.cuda.Event()
+        # Don't overwrite buffers the compute stream may still be reading.
+        self._h2d_stream.wait_event(st.consumed_event)
+        with torch.cuda.stream(self._h2d_stream):
+            if st.staged is None:
+                st.staged = {
+                    k: torch.empty(t.shape, dtype=t.dtype, device=self.cfg.device)
+                    for k, t in cpu_out.items()
+                }
+            for k, t in cpu_out.items():
+                st.staged[k].copy_(t, non_blocking=True)
+            st.h2d_event.record()
+        st.prefetched = cpu_out
+
     def act_once(self):
         cfg = self.cfg
         cur = self.next_env_index
         self.next_env_index = (self.next_env_index + 1) % cfg.num_actor_batches
         env_state = self.env_states[cur]
         t0 = time.perf_counter() if self.profile else 0.0
-        if env_state.future is None:
+        if env_state.future is None and env_state.prefetched is None:
             env_state.future = self.envs.step(cur, env_state.prev_action)
-        cpu_env_outputs = env_state.future.result()
+        prefetched = env_state.prefetched is not None
+        if prefetched:
+            cpu_env_outputs = env_state.prefetched
+            env_state.prefetched = None
+        else:
+            cpu_env_outputs = env_state.future.result()
         t0 = self._t("act_env_wait", t0)
         stream_ctx = (
             torch.cuda.stream(self.actor_stream)
@@ -550,11 +609,15 @@ class ImpalaPeer:
             else contextlib.nullcontext()
         )
         with stream_ctx:
-            self._act_body(cfg, cur, env_state, cpu_env_outputs, t0)
+            self._act_body(cfg, cur, env_state, cpu_env_outputs, t0, prefetched)
 
-    def _act_body(self, cfg, cur, env_state, cpu_env_outputs, t0):
+    def _act_body(self, cfg, cur, env_state, cpu_env_outputs, t0, prefetched=False):
 
-        if self._shm_registered:
+        if prefetched:
+            # results already live in the staged device buffers (copy stream)
+            torch.cuda.current_stream().wait_event(env_state.h2d_event)
+            env_outputs = dict(env_state.staged)
+        elif self._shm_registered:
             # shm region is hipHostRegistered: direct async DMA, no bounce.
             env_outputs = {
                 k: t.to(cfg.device, non_blocking=True) for k, t in cpu_env_outputs.items()
@@ -613,4 +676,8 @@ class ImpalaPeer:
             # Carry the last entry of the previous unroll into the next one.
             env_state.initial_core_state = prev_core_state
             env_state.time_batcher.stack(last_data)
+        if prefetched:
+            # all reads of the staged buffers are now issued on this stream;
+            # the next prefetch overwrite waits on this event
+            env_state.consumed_event.record()
         self._t("act_batch", t0)
