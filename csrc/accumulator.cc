@@ -589,12 +589,24 @@ void Accumulator::reduceGradients(int64_t batchSize) {
     throw RpcError("reduce_gradients called when wants_gradients() is false");
   }
   NoGrad ng;
+  // Multi-tensor: one launch folds every param grad into its bucket slice
+  // and one zeroes them (vs 2 launches per param — 40 tiny latency-bound
+  // kernels each on the IMPALA model).
+  std::vector<at::Tensor> slices, flatGrads, grads;
+  slices.reserve(params_.size());
+  flatGrads.reserve(params_.size());
+  grads.reserve(params_.size());
   for (size_t i = 0; i < params_.size(); ++i) {
     auto& p = params_[i];
     if (p.grad().defined()) {
-      s.flat.narrow(0, offsets_[i], numels_[i]).add_(p.grad().flatten());
-      p.grad().zero_();
+      slices.push_back(s.flat.narrow(0, offsets_[i], numels_[i]));
+      flatGrads.push_back(p.grad().flatten());  // view for contiguous grads
+      grads.push_back(p.grad());                // originals for the zero
     }
+  }
+  if (!grads.empty()) {
+    at::_foreach_add_(slices, flatGrads);
+    at::_foreach_zero_(grads);
   }
   decided_ = true;
   s.newBatch += batchSize;
@@ -604,9 +616,11 @@ void Accumulator::reduceGradients(int64_t batchSize) {
 void Accumulator::zeroGradients() {
   std::lock_guard<std::mutex> lk(mu_);
   NoGrad ng;
+  std::vector<at::Tensor> grads;
   for (auto& p : params_) {
-    if (p.grad().defined()) p.grad().zero_();
+    if (p.grad().defined()) grads.push_back(p.grad());
   }
+  if (!grads.empty()) at::_foreach_zero_(grads);
   hasGradients_ = false;
   applyPendingLocked();
 }

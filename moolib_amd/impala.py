@@ -194,6 +194,9 @@ class ImpalaPeer:
             lr=cfg.learning_rate,
             betas=(cfg.adam_beta1, cfg.adam_beta2),
             eps=cfg.adam_eps,
+            # multi-tensor kernels: one launch per op across all 40 params
+            # instead of 40 tiny latency-bound launches each
+            foreach=True,
         )
         if cfg.lr_schedule:
             factor = cfg.unroll_length * cfg.virtual_batch_size / cfg.total_steps
@@ -353,10 +356,12 @@ class ImpalaPeer:
         if not self.bf16_shadow:
             return
         with torch.no_grad():
-            for pf, pb in zip(self._master_params, self._fwd_params):
-                pb.copy_(pf, non_blocking=True)
-            for bf, bb in zip(self.model.buffers(), self.fwd_model.buffers()):
-                bb.copy_(bf, non_blocking=True)
+            # one multi-tensor cast-copy launch for all params (fp32->bf16)
+            torch._foreach_copy_(self._fwd_params, self._master_params)
+            bufs_dst = list(self.fwd_model.buffers())
+            bufs_src = list(self.model.buffers())
+            if bufs_dst:
+                torch._foreach_copy_(bufs_dst, bufs_src)
 
     def _learn_fn(self, data):
         """Forward + V-trace + fused loss + backward. hipGraph-capturable:
@@ -423,13 +428,18 @@ class ImpalaPeer:
         self._learn_call(data)
         if self.bf16_shadow:
             with torch.no_grad():
+                dsts, srcs = [], []
                 for pf, pb in zip(self._master_params, self._fwd_params):
                     if pb.grad is None:
                         continue
                     if pf.grad is None:
                         pf.grad = pb.grad.float()
                     else:
-                        pf.grad.copy_(pb.grad, non_blocking=True)
+                        dsts.append(pf.grad)
+                        srcs.append(pb.grad)
+                if dsts:
+                    # one multi-tensor bf16->fp32 cast-copy launch
+                    torch._foreach_copy_(dsts, srcs)
         self.stats["env_train_steps"] += self.cfg.unroll_length * self.cfg.batch_size
 
     def _actor_fn(self, inputs):
