@@ -338,6 +338,7 @@ __global__ void conv1_u8_nhwc_kernel(const uint8_t* __restrict__ in,   // [N,4,H
 }  // namespace
 
 #include "conv3x3.hip.inc"
+#include "wgrad3x3.hip.inc"
 #include "lstm.hip.inc"
 
 // ------------------------------------------------------------ wrappers
@@ -863,6 +864,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w_packed"), py::arg("k"), py::arg("relu_in") = false,
         py::arg("bias_in") = py::none(), py::arg("epi") = 0, py::arg("bias1") = py::none(),
         py::arg("res") = py::none(), py::arg("bias2") = py::none(), py::arg("rt") = 0);
+  m.def("wgrad3x3_nhwc", &wgrad3x3_nhwc,
+        "3x3/s1/p1 conv weight gradient on MFMA (row-slab LDS staging, gfx950)");
   m.def("lstm_fused_fwd", &lstm_fused_fwd, "fused masked LSTM sequence scan fwd (MFMA, gfx950)");
   m.def("lstm_fused_bwd", &lstm_fused_bwd, "fused masked LSTM sequence scan bwd (MFMA, gfx950)");
   m.def("vtrace_from_log_rhos", &vtrace_from_log_rhos, "fused V-trace scan (gfx950)");

@@ -788,3 +788,37 @@ class TestConv3x3Autograd:
         for n, p in model.named_parameters():
             assert p.grad is not None, n
             assert torch.isfinite(p.grad.float()).all(), n
+
+
+@gpu
+@requires_gpu
+class TestWgradKernel:
+    def test_matches_fp32_reference(self):
+        """wgrad3x3_nhwc vs aten.convolution_backward in fp32, across
+        shapes with W%32 tails, halo rows, and both C/K combos."""
+        from moolib_amd import _kernels
+
+        torch.manual_seed(9)
+        for N, C, H, W, K in [
+            (3, 16, 42, 42, 16),
+            (2, 32, 21, 21, 32),
+            (2, 32, 11, 11, 32),
+            (1, 16, 84, 84, 32),
+            (2, 16, 5, 7, 16),
+        ]:
+            x = torch.randn(N, C, H, W, device="cuda", dtype=torch.bfloat16).contiguous(
+                memory_format=torch.channels_last
+            )
+            dy = torch.randn(N, K, H, W, device="cuda", dtype=torch.bfloat16).contiguous(
+                memory_format=torch.channels_last
+            )
+            g = _kernels.wgrad3x3_nhwc(x, dy)[: 9 * C]
+            dw = g.view(3, 3, C, K).permute(3, 2, 0, 1)
+            w = torch.zeros(K, C, 3, 3, device="cuda")
+            ref = torch.ops.aten.convolution_backward(
+                dy.float(), x.float(), w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                [False, True, False],
+            )[1]
+            err = (dw - ref).abs().max().item()
+            scale = max(ref.abs().max().item(), 1.0)
+            assert err / scale < 0.02, ((N, C, H, W, K), err, scale)
