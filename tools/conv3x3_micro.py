@@ -66,5 +66,35 @@ def main():
         )
 
 
+def wgrad_micro():
+    from moolib_amd import _kernels
+
+    print("\nwgrad (dW) micro:")
+    print(f"{'shape':<18} {'C->K':>6} {'ours_ms':>8} {'miopen_ms':>10}  ratio")
+    for N, C, H, W, K in [
+        (672, 16, 42, 42, 16),
+        (672, 16, 42, 42, 32),
+        (672, 32, 21, 21, 32),
+        (672, 32, 11, 11, 32),
+    ]:
+        x = torch.randn(N, C, H, W, device="cuda", dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last
+        )
+        dy = torch.randn(N, K, H, W, device="cuda", dtype=torch.bfloat16).contiguous(
+            memory_format=torch.channels_last
+        )
+        w = torch.zeros(K, C, 3, 3, device="cuda", dtype=torch.bfloat16)
+        t_ours = timeit(lambda: _kernels.wgrad3x3_nhwc(x, dy))
+        t_mi = timeit(
+            lambda: torch.ops.aten.convolution_backward(
+                dy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                [False, True, False],
+            )[1]
+        )
+        print(f"[{N},{C},{H},{W}]".ljust(18) + f" {C}->{K}".rjust(6)
+              + f" {t_ours:>8.3f} {t_mi:>10.3f}  {t_ours/t_mi:.2f}")
+
+
 if __name__ == "__main__":
     main()
+    wgrad_micro()
