@@ -149,16 +149,22 @@ class _Conv3x3Fn(torch.autograd.Function):
             if dx.dtype != ctx.x_dtype:
                 dx = dx.to(ctx.x_dtype)
         if ctx.needs_input_grad[1]:
-            if os.environ.get("MOOLIB_AMD_NO_WGRAD_KERNEL"):
-                dw = torch.ops.aten.convolution_backward(
-                    dy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
-                    [False, True, False],
-                )[1]
-            else:
+            if os.environ.get("MOOLIB_AMD_WGRAD_KERNEL"):
+                # Opt-in: our row-slab wgrad kernel is numerics-correct
+                # (TestWgradKernel) but 3-5x slower than MIOpen's igemm_wrw
+                # at IMPALA shapes (measured r2: per-row staging+sync
+                # overhead dwarfs the ~10 MFMAs/wave a 21-42px row yields;
+                # a multi-row-slab rewrite is the known fix). MIOpen keeps
+                # the weight gradient by default.
                 K, C = w.shape[0], w.shape[1]
                 # [9C-padded, K] fp32 GEMM gradient -> [K, C, 3, 3]
                 g = _k().wgrad3x3_nhwc(x, dy)[: 9 * C]
                 dw = g.view(3, 3, C, K).permute(3, 2, 0, 1)
+            else:
+                dw = torch.ops.aten.convolution_backward(
+                    dy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                    [False, True, False],
+                )[1]
             if dw.dtype != w.dtype:  # bf16 weights (shadow mode) / fp32 master
                 dw = dw.to(w.dtype)
         return dx, dw, None
