@@ -370,10 +370,8 @@ class ImpalaPeer:
         model = self.fwd_model
         if self._learn_rng is not None:
             model.sample_generator = self._learn_rng
-        if self.bf16_shadow:
-            for pb in self._fwd_params:
-                if pb.grad is not None:
-                    pb.grad.zero_()
+        # (shadow-grad zeroing happens in compute_gradients, outside any
+        # captured region, as ONE foreach launch)
         env_outputs = data["env_outputs"]
         actor_outputs = data["actor_outputs"]
         initial_core_state = data["initial_core_state"]
