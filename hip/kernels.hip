@@ -20,6 +20,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 
+#include <cstdlib>
 #include <vector>
 
 #define DEV_INLINE __device__ __forceinline__
