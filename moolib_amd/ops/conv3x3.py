@@ -150,12 +150,13 @@ class _Conv3x3Fn(torch.autograd.Function):
                 dx = dx.to(ctx.x_dtype)
         if ctx.needs_input_grad[1]:
             if os.environ.get("MOOLIB_AMD_WGRAD_KERNEL"):
-                # Opt-in: our row-slab wgrad kernel is numerics-correct
-                # (TestWgradKernel) but 3-5x slower than MIOpen's igemm_wrw
-                # at IMPALA shapes (measured r2: per-row staging+sync
-                # overhead dwarfs the ~10 MFMAs/wave a 21-42px row yields;
-                # a multi-row-slab rewrite is the known fix). MIOpen keeps
-                # the weight gradient by default.
+                # Opt-in: the row-slab wgrad kernel is numerics-correct
+                # (TestWgradKernel) and, after the accumulator-scratch fix,
+                # within 1.1-1.25x of MIOpen's igemm_wrw at IMPALA learner
+                # shapes (102us vs 93us, 69us vs 56us —
+                # gpurun_out/r2l_sweep.txt). MIOpen keeps the default until
+                # an A/B shows parity; the remaining gap is the per-slab
+                # atomic fold.
                 K, C = w.shape[0], w.shape[1]
                 # [9C-padded, K] fp32 GEMM gradient -> [K, C, 3, 3]
                 g = _k().wgrad3x3_nhwc(x, dy)[: 9 * C]
