@@ -174,19 +174,26 @@ class AtariNet(nn.Module):
         first = 0
         if (
             kernels is not None
-            and not torch.is_grad_enabled()  # actor path; learner conv needs autograd
             and self.sections[0].conv.in_channels == 4
             and not _os.environ.get("MOOLIB_AMD_NO_CONV1_KERNEL")
         ):
             # Fused uint8 frames -> conv1 + bias in one kernel (no separate
-            # preprocessing pass, no 19 MB fp intermediate).
-            c1 = self.sections[0].conv
-            w = c1.weight.detach().to(torch.bfloat16).permute(2, 3, 1, 0).contiguous()
-            b = c1.bias.detach().to(torch.bfloat16)
-            x = kernels.conv1_u8_nhwc(x, w, b, 1.0 / 255.0)
+            # preprocessing pass, no 19 MB fp intermediate). Under grad the
+            # same kernel runs through an autograd Function whose backward
+            # is our wgrad kernel + a bias-sum — the first layer never
+            # touches MIOpen in either direction.
+            sec0 = self.sections[0]
+            c1 = sec0.conv
+            if torch.is_grad_enabled() and c1.weight.requires_grad:
+                from moolib_amd.ops import conv3x3 as c3m
+
+                x = c3m.conv1_u8_autograd(x, c1, 1.0 / 255.0)
+            else:
+                w = c1.weight.detach().to(torch.bfloat16).permute(2, 3, 1, 0).contiguous()
+                b = c1.bias.detach().to(torch.bfloat16)
+                x = kernels.conv1_u8_nhwc(x, w, b, 1.0 / 255.0)
             from moolib_amd.ops.pool import maxpool3x3s2
 
-            sec0 = self.sections[0]
             x = maxpool3x3s2(x)
             from moolib_amd.ops import fused_bias
 

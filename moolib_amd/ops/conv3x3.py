@@ -176,6 +176,38 @@ def conv3x3_autograd(x, conv_module):
     return _Conv3x3Fn.apply(x, conv_module.weight, conv_module)
 
 
+class _Conv1U8Fn(torch.autograd.Function):
+    """First conv (uint8 frames -> 16ch) with fwd on the fused conv1_u8
+    kernel and dW/db on our wgrad kernel — no MIOpen anywhere in the layer
+    (frames never need an input gradient)."""
+
+    @staticmethod
+    def forward(ctx, x_u8, weight, bias, scale):
+        k = _k()
+        w = weight.detach().to(torch.bfloat16).permute(2, 3, 1, 0).contiguous()
+        b = bias.detach().to(torch.bfloat16)
+        y = k.conv1_u8_nhwc(x_u8, w, b, scale)
+        ctx.save_for_backward(x_u8, weight)
+        ctx.scale = scale
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x_u8, w = ctx.saved_tensors
+        k = _k()
+        dyb = dy.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+        xb = k.frames_u8_to_bf16_nhwc(x_u8, ctx.scale)  # same scaling as fwd
+        K, C = w.shape[0], w.shape[1]
+        g = k.wgrad3x3_nhwc(xb, dyb)[: 9 * C]
+        dw = g.view(3, 3, C, K).permute(3, 2, 0, 1).to(w.dtype)
+        db = dy.float().sum((0, 2, 3)).to(w.dtype)
+        return None, dw, db, None
+
+
+def conv1_u8_autograd(x_u8, conv_module, scale):
+    return _Conv1U8Fn.apply(x_u8, conv_module.weight, conv_module.bias, scale)
+
+
 def conv3x3(x, w_packed, k, relu_in=False, bias_in=None, epi=EPI_NONE,
             bias1=None, res=None, bias2=None, rt=0):
     """out = conv3x3_s1_p1( relu(x + bias_in) if relu_in else x ) then

@@ -822,3 +822,33 @@ class TestWgradKernel:
             err = (dw - ref).abs().max().item()
             scale = max(ref.abs().max().item(), 1.0)
             assert err / scale < 0.02, ((N, C, H, W, K), err, scale)
+
+
+@gpu
+@requires_gpu
+class TestConv1Autograd:
+    def test_first_layer_grads_match_fp32(self):
+        """conv1_u8_autograd: fused fwd + wgrad-kernel dW + bias-sum db vs
+        the fp32 reference (frames/255 -> conv2d)."""
+        import torch.nn.functional as F
+
+        from moolib_amd.ops import conv3x3 as c3
+
+        torch.manual_seed(12)
+        conv = torch.nn.Conv2d(4, 16, 3, padding=1).to("cuda").to(torch.bfloat16)
+        x = torch.randint(0, 256, (6, 4, 84, 84), dtype=torch.uint8, device="cuda")
+        y = c3.conv1_u8_autograd(x, conv, 1.0 / 255.0)
+        g = torch.randn_like(y)
+        y.backward(g)
+        dw, db = conv.weight.grad.float(), conv.bias.grad.float()
+
+        xf = (x.float() / 255.0).requires_grad_(False)
+        wf = conv.weight.detach().float().clone().requires_grad_()
+        bf = conv.bias.detach().float().clone().requires_grad_()
+        yf = F.conv2d(xf, wf, bf, padding=1)
+        yf.backward(g.float())
+
+        for name, got, want in [("y", y.float(), yf.detach()), ("dw", dw, wf.grad), ("db", db, bf.grad)]:
+            err = (got - want).abs().max().item()
+            scale = max(want.abs().max().item(), 1.0)
+            assert err / scale < 0.03, (name, err, scale)
