@@ -21,6 +21,17 @@ ROOT = Path(__file__).resolve().parent
 CORE_SOURCES = sorted(str(p) for p in (ROOT / "csrc").glob("*.cc"))
 HIP_SOURCES = sorted(str(p) for p in (ROOT / "hip").glob("*.hip"))
 
+# The hipcc compile step emits no dependency file, so ninja cannot see the
+# *.hip.inc includes — an edit there would silently ship a STALE kernels.so
+# (bit us in round 2). Bump the including .hip's mtime whenever an include
+# is newer.
+for hip in HIP_SOURCES:
+    hp = Path(hip)
+    for inc in (ROOT / "hip").glob("*.hip.inc"):
+        if inc.stat().st_mtime > hp.stat().st_mtime:
+            hp.touch()
+            break
+
 common_args = [
     "-O2",
     "-std=c++17",
