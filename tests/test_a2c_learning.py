@@ -12,8 +12,8 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", "examples"))
 
 
 @pytest.mark.timeout(600)
-def test_a2c_cartpole_learns(tmp_path):
-    os.chdir(tmp_path)  # logs.tsv etc. go to tmp
+def test_a2c_cartpole_learns(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)  # logs.tsv etc. go to tmp; RESTORED after
     import a2c
 
     # Two seeds: RL on a tiny budget has real variance; requiring one of two

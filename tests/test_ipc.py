@@ -201,7 +201,9 @@ t = c.sync("h", "get")
 assert torch.equal(t.cpu(), x.cpu())
 print("OK")
 """
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     env = dict(os.environ, MOOLIB_AMD_NO_IPC_RPC="1")
+    env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
     r = subprocess.run([sys.executable, "-c", code], capture_output=True, text=True,
-                       timeout=110, env=env)
+                       timeout=110, env=env, cwd=repo)
     assert r.returncode == 0 and "OK" in r.stdout, r.stderr[-2000:]

@@ -78,12 +78,14 @@ t[0] = 1.0           # byte copy: host must NOT see the write
 assert buf[0] == 7.0
 print("OK")
 """
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
         env = dict(os.environ, MOOLIB_AMD_NO_IPC_RPC="1")
+        env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
         last = None
         for _ in range(2):  # one retry: the subprocess cold-imports torch
             r = subprocess.run(
                 [sys.executable, "-c", code], capture_output=True, text=True,
-                timeout=240, env=env,
+                timeout=240, env=env, cwd=repo,
             )
             if r.returncode == 0 and "OK" in r.stdout:
                 return
