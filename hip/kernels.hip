@@ -848,9 +848,121 @@ void batched_copy(std::vector<at::Tensor> dsts, std::vector<at::Tensor> srcs) {
   flush();
 }
 
+// ------------------------------------------- batched conv-weight repack
+//
+// The conv3x3 path keeps per-conv MFMA-packed weight buffers (fwd pack +
+// dgrad pack) that must refresh once per optimizer step. Doing that with
+// torch ops is ~14 kernels per conv per direction (flip/permute/cat/
+// contiguous) replayed inside the captured optimizer graph (~0.4 ms and
+// ~200 launches per step in the r2 profile); this kernel rewrites EVERY
+// pack in one launch: each output element decodes its (ct, kk, lane, e)
+// fragment coordinate and reads the one weight element it mirrors.
+// Layouts mirror ops/lstm.py pack_mfma_b and ops/conv3x3.py
+// pack_weight/pack_weight_dgrad exactly (tested against them).
+
+namespace repack3x3 {
+
+using bf16 = __bf16;
+
+struct Desc {
+  const bf16* w;  // [K, C, 3, 3], arbitrary strides (channels_last safe)
+  bf16* outF;     // fwd pack, ceil(9C/32)*32*K elems, or null
+  bf16* outD;     // dgrad pack, ceil(9K/32)*32*C elems, or null
+  int64_t sK, sC, sH, sW;  // element strides of w
+  int32_t C, K;
+};
+constexpr int kMaxDescs = 20;
+struct Pack {
+  Desc d[kMaxDescs];
+  int32_t n;
+};
+
+__global__ void repack3x3_kernel(Pack pack) {
+  const Desc& d = pack.d[blockIdx.y];
+  const int KKf = (9 * d.C + 31) / 32;
+  const int KKd = (9 * d.K + 31) / 32;
+  const int nF = d.outF ? KKf * 32 * d.K : 0;
+  const int nD = d.outD ? KKd * 32 * d.C : 0;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < nF + nD;
+       i += gridDim.x * blockDim.x) {
+    const bool isF = i < nF;
+    const int idx = isF ? i : i - nF;
+    const int KK = isF ? KKf : KKd;
+    const int CC = isF ? d.C : d.K;  // in-channels of this direction
+    // idx = ((ct*KK + kk)*64 + lane)*8 + e
+    const int e = idx & 7;
+    const int lane = (idx >> 3) & 63;
+    const int rest = idx >> 9;
+    const int kk = rest % KK;
+    const int ct = rest / KK;
+    const int kd = kk * 32 + (lane >> 4) * 8 + e;
+    const int n = ct * 16 + (lane & 15);
+    bf16 v = (bf16)0.f;
+    if (kd < 9 * CC) {
+      const int tap = kd / CC, c = kd % CC;
+      const int kh = tap / 3, kw = tap % 3;
+      if (isF) {
+        v = d.w[n * d.sK + c * d.sC + kh * d.sH + kw * d.sW];
+      } else {
+        // dgrad: W'[c_out=n... ] = w[c][n][2-kh][2-kw]
+        v = d.w[c * d.sK + n * d.sC + (2 - kh) * d.sH + (2 - kw) * d.sW];
+      }
+    }
+    (isF ? d.outF : d.outD)[idx] = v;
+  }
+}
+
+}  // namespace repack3x3
+
+// Refresh fwd/dgrad MFMA packs for a batch of convs in ONE launch.
+// dgrad entries may be empty tensors (conv1 has no dgrad pack).
+void repack3x3_batched(std::vector<at::Tensor> ws, std::vector<at::Tensor> fwdBufs,
+                       std::vector<at::Tensor> dgradBufs) {
+  TORCH_CHECK(ws.size() == fwdBufs.size() && ws.size() == dgradBufs.size(),
+              "repack3x3: list length mismatch");
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  repack3x3::Pack pack;
+  pack.n = 0;
+  auto flush = [&]() {
+    if (pack.n == 0) return;
+    hipLaunchKernelGGL(repack3x3::repack3x3_kernel, dim3(32, pack.n), dim3(256), 0, stream,
+                       pack);
+    pack.n = 0;
+  };
+  for (size_t i = 0; i < ws.size(); ++i) {
+    at::Tensor& w = ws[i];
+    TORCH_CHECK(w.is_cuda() && w.dim() == 4 && w.scalar_type() == at::kBFloat16,
+                "repack3x3: bf16 4D CUDA weight expected");
+    const int K = w.size(0), C = w.size(1);
+    repack3x3::Desc e;
+    e.w = reinterpret_cast<const repack3x3::bf16*>(w.data_ptr());
+    e.sK = w.stride(0);
+    e.sC = w.stride(1);
+    e.sH = w.stride(2);
+    e.sW = w.stride(3);
+    e.C = C;
+    e.K = K;
+    e.outF = nullptr;
+    e.outD = nullptr;
+    if (fwdBufs[i].defined() && fwdBufs[i].numel() > 0) {
+      TORCH_CHECK(fwdBufs[i].numel() == (9 * C + 31) / 32 * 32 * K, "repack3x3: fwd size");
+      e.outF = reinterpret_cast<repack3x3::bf16*>(fwdBufs[i].data_ptr());
+    }
+    if (dgradBufs[i].defined() && dgradBufs[i].numel() > 0) {
+      TORCH_CHECK(dgradBufs[i].numel() == (9 * K + 31) / 32 * 32 * C, "repack3x3: dgrad size");
+      e.outD = reinterpret_cast<repack3x3::bf16*>(dgradBufs[i].data_ptr());
+    }
+    pack.d[pack.n++] = e;
+    if (pack.n == repack3x3::kMaxDescs) flush();
+  }
+  flush();
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "moolib_amd gfx950 HIP kernels";
   m.def("register_host_memory", &register_host_memory);
+  m.def("repack3x3_batched", &repack3x3_batched,
+        "refresh all conv MFMA weight packs (fwd + dgrad) in one launch");
   m.def("batched_copy", &batched_copy,
         "copy many (dst<-src) slice pairs in one kernel launch (gfx950)");
   m.def("maxpool3x3s2_fwd", &maxpool3x3s2_fwd, "NHWC 3x3/2 maxpool forward (gfx950)");

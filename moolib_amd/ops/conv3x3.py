@@ -83,19 +83,43 @@ def packed_buffer(conv):
 
 def repack(module):
     """Refresh every cached packed weight in `module` (call after the
-    optimizer step; safe inside a hipGraph capture)."""
+    optimizer step; safe inside a hipGraph capture).
+
+    On GPU with bf16 weights this is ONE kernel launch for all convs and
+    both directions (repack3x3_batched); the torch-op path (flip/permute/
+    cat per conv, ~14 kernels each) remains the fallback."""
     import torch.nn as nn
 
+    ws, fs, ds, caches = [], [], [], []
+    fallback = []
     for m in module.modules():
         if isinstance(m, nn.Conv2d):
             cache = getattr(m, "_c3_cache", None)
-            if cache is not None:
-                cache[1].copy_(pack_weight(m.weight.detach()))
-                cache[0] = m.weight._version
             dcache = getattr(m, "_c3_dgrad_cache", None)
+            if cache is None and dcache is None:
+                continue
+            w = m.weight.detach()
+            if w.is_cuda and w.dtype == torch.bfloat16:
+                ws.append(w)
+                fs.append(cache[1] if cache is not None else w.new_empty(0))
+                ds.append(dcache[1] if dcache is not None else w.new_empty(0))
+                caches.append((m, cache, dcache))
+            else:
+                fallback.append((m, cache, dcache))
+    if ws:
+        _k().repack3x3_batched(ws, fs, ds)
+        for m, cache, dcache in caches:
+            if cache is not None:
+                cache[0] = m.weight._version
             if dcache is not None:
-                dcache[1].copy_(pack_weight_dgrad(m.weight.detach()))
                 dcache[0] = m.weight._version
+    for m, cache, dcache in fallback:
+        if cache is not None:
+            cache[1].copy_(pack_weight(m.weight.detach()))
+            cache[0] = m.weight._version
+        if dcache is not None:
+            dcache[1].copy_(pack_weight_dgrad(m.weight.detach()))
+            dcache[0] = m.weight._version
 
 
 def pack_weight_dgrad(w):

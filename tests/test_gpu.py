@@ -852,3 +852,34 @@ class TestConv1Autograd:
             err = (got - want).abs().max().item()
             scale = max(want.abs().max().item(), 1.0)
             assert err / scale < 0.03, (name, err, scale)
+
+
+@gpu
+@requires_gpu
+class TestBatchedRepack:
+    def test_matches_torch_pack(self):
+        """repack3x3_batched must reproduce pack_weight AND
+        pack_weight_dgrad bit-exactly, incl. channels_last weights and the
+        C=4 first conv (no dgrad pack)."""
+        from moolib_amd import _kernels
+        from moolib_amd.ops import conv3x3 as c3
+
+        torch.manual_seed(21)
+        specs = [(16, 16), (16, 32), (32, 32), (4, 16)]
+        ws, fs, ds, refs = [], [], [], []
+        for C, K in specs:
+            w = torch.randn(K, C, 3, 3, device="cuda", dtype=torch.bfloat16)
+            if C != 4:
+                w = w.to(memory_format=torch.channels_last)
+            f_ref = c3.pack_weight(w)
+            d_ref = c3.pack_weight_dgrad(w) if C != 4 else None
+            ws.append(w)
+            fs.append(torch.empty_like(f_ref))
+            ds.append(torch.empty_like(d_ref) if d_ref is not None else w.new_empty(0))
+            refs.append((f_ref, d_ref))
+        _kernels.repack3x3_batched(ws, fs, ds)
+        torch.cuda.synchronize()
+        for (C, K), fbuf, dbuf, (f_ref, d_ref) in zip(specs, fs, ds, refs):
+            assert torch.equal(fbuf, f_ref), (C, K, "fwd")
+            if d_ref is not None:
+                assert torch.equal(dbuf, d_ref), (C, K, "dgrad")
