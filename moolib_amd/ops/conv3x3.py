@@ -31,7 +31,7 @@ def available(C, K):
     # version-checked persistent buffers (repack() once per optimizer step;
     # the r1 regression was per-replay repacking inside the captured
     # forward), same-box interleaved A/B measured the kernel path ~4%
-    # faster end-to-end (gpurun_out/r2_{on,off}*.json: 60.7k vs 58.4k and
+    # faster end-to-end (profiles/evidence/r2_{on,off}*.json: 60.7k vs 58.4k and
     # 58.7k vs 56.2k f/s). MOOLIB_AMD_CONV3_KERNEL=0 reverts to MIOpen.
     if os.environ.get("MOOLIB_AMD_CONV3_KERNEL", "1") == "0":
         return False
@@ -178,7 +178,7 @@ class _Conv3x3Fn(torch.autograd.Function):
                 # (TestWgradKernel) and, after the accumulator-scratch fix,
                 # within 1.1-1.25x of MIOpen's igemm_wrw at IMPALA learner
                 # shapes (102us vs 93us, 69us vs 56us —
-                # gpurun_out/r2l_sweep.txt). MIOpen keeps the default until
+                # profiles/evidence/r2l_sweep.txt). MIOpen keeps the default until
                 # an A/B shows parity; the remaining gap is the per-slab
                 # atomic fold.
                 K, C = w.shape[0], w.shape[1]
