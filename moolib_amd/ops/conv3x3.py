@@ -1,10 +1,13 @@
-"""Fused MFMA 3x3 NHWC conv for the actor inference path.
+"""Fused MFMA 3x3 NHWC conv (actor inference AND learner autograd paths).
 
 Wraps hip/conv3x3.hip.inc: implicit-GEMM conv on mfma_f32_16x16x32_bf16
 with the producer relu (+pending bias) fused into the input load and the
 bias / bias+relu / bias+residual epilogue fused into the store. An IMPALA
 residual block (reference examples/atari/models.py:30-42) becomes two
-kernel launches. Forward-only — gate on `not torch.is_grad_enabled()`.
+kernel launches on the actor. The learner uses `conv3x3_autograd`
+(forward + input-gradient on the same kernel; weight gradient via MIOpen
+wrw by default, our wgrad3x3 kernel opt-in) and `conv1_u8_autograd` for
+the uint8 first layer.
 """
 import os
 
@@ -149,8 +152,8 @@ def dgrad_buffer(conv):
 class _Conv3x3Fn(torch.autograd.Function):
     """3x3/s1/p1 NHWC conv with fwd AND dgrad on the MFMA kernel.
 
-    Weight gradient goes through aten.convolution_backward (MIOpen wrw
-    igemm) — the one learner conv pass without a hand-written kernel yet.
+    Weight gradient defaults to aten.convolution_backward (MIOpen wrw
+    igemm); MOOLIB_AMD_WGRAD_KERNEL=1 switches to our wgrad3x3 kernel.
     Drop-in for F.conv2d(x, w, None, padding=1) on supported shapes."""
 
     @staticmethod
