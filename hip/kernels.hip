@@ -302,38 +302,65 @@ __global__ void conv1_u8_nhwc_kernel(const uint8_t* __restrict__ in,   // [N,4,H
   for (int i = threadIdx.x; i < CO; i += blockDim.x) bsm[i] = (float)bias[i];
   __syncthreads();
 
+  // 2 output pixels per thread: each (kh, ci) input row contributes a
+  // 4-byte span (iw-1..iw+2) fetched as ONE dword in the interior — 12
+  // dword loads replace the naive 72 scalar byte loads per 2 pixels.
+  const int WT = (W + 1) / 2;  // width tiles
   int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t total = (int64_t)N * H * W;
+  int64_t total = (int64_t)N * H * WT;
   if (tid >= total) return;
-  int ow = tid % W;
-  int64_t t = tid / W;
+  int ow0 = (int)(tid % WT) * 2;
+  int64_t t = tid / WT;
   int oh = t % H;
   int n = t / H;
 
-  float acc[CO];
+  float acc[2][CO];
 #pragma unroll
-  for (int co = 0; co < CO; ++co) acc[co] = bsm[co];
+  for (int px = 0; px < 2; ++px)
+#pragma unroll
+    for (int co = 0; co < CO; ++co) acc[px][co] = bsm[co];
 
 #pragma unroll
   for (int kh = 0; kh < 3; ++kh) {
     int ih = oh + kh - 1;
     if (ih < 0 || ih >= H) continue;
 #pragma unroll
-    for (int kw = 0; kw < 3; ++kw) {
-      int iw = ow + kw - 1;
-      if (iw < 0 || iw >= W) continue;
-      const float* wp = &wsm[(kh * 3 + kw) * CI * CO];
+    for (int ci = 0; ci < CI; ++ci) {
+      const uint8_t* row = in + (((int64_t)n * CI + ci) * H + ih) * W;
+      float xv[4];  // input columns ow0-1 .. ow0+2
+      if (ow0 >= 1 && ow0 + 2 < W) {
+        uint32_t v;
+        __builtin_memcpy(&v, row + ow0 - 1, 4);
 #pragma unroll
-      for (int ci = 0; ci < CI; ++ci) {
-        float xv = (float)in[(((int64_t)n * CI + ci) * H + ih) * W + iw] * scale;
+        for (int j = 0; j < 4; ++j) xv[j] = (float)((v >> (8 * j)) & 0xff) * scale;
+      } else {
 #pragma unroll
-        for (int co = 0; co < CO; ++co) acc[co] = fmaf(xv, wp[ci * CO + co], acc[co]);
+        for (int j = 0; j < 4; ++j) {
+          int iw = ow0 - 1 + j;
+          xv[j] = (iw >= 0 && iw < W) ? (float)row[iw] * scale : 0.f;
+        }
+      }
+#pragma unroll
+      for (int kw = 0; kw < 3; ++kw) {
+        const float* wp = &wsm[((kh * 3 + kw) * CI + ci) * CO];
+#pragma unroll
+        for (int px = 0; px < 2; ++px) {
+#pragma unroll
+          for (int co = 0; co < CO; ++co) {
+            acc[px][co] = fmaf(xv[kw + px], wp[co], acc[px][co]);
+          }
+        }
       }
     }
   }
-  hip_bfloat16* op = out + tid * CO;
 #pragma unroll
-  for (int co = 0; co < CO; ++co) op[co] = (hip_bfloat16)acc[co];
+  for (int px = 0; px < 2; ++px) {
+    const int ow = ow0 + px;
+    if (ow >= W) break;
+    hip_bfloat16* op = out + (((int64_t)n * H + oh) * W + ow) * CO;
+#pragma unroll
+    for (int co = 0; co < CO; ++co) op[co] = (hip_bfloat16)acc[px][co];
+  }
 }
 
 }  // namespace
@@ -478,7 +505,7 @@ at::Tensor conv1_u8_nhwc(at::Tensor x, at::Tensor w, at::Tensor bias, double sca
   int N = xc.size(0), H = xc.size(2), W = xc.size(3);
   auto out = at::empty({N, 16, H, W},
                        xc.options().dtype(at::kBFloat16).memory_format(at::MemoryFormat::ChannelsLast));
-  int64_t total = (int64_t)N * H * W;
+  int64_t total = (int64_t)N * H * ((W + 1) / 2);  // 2 output pixels/thread
   int threads = 256;
   int64_t blocks = (total + threads - 1) / threads;
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
