@@ -189,8 +189,11 @@ class _Conv3x3Fn(torch.autograd.Function):
                 g = _k().wgrad3x3_nhwc(x, dy)[: 9 * C]
                 dw = g.view(3, 3, C, K).permute(3, 2, 0, 1)
             else:
+                # wrw requires x/dy/w in one dtype; under plain autocast the
+                # master weights are fp32 while the saved x and dy are bf16
+                wb = w if w.dtype == dy.dtype else w.to(dy.dtype)
                 dw = torch.ops.aten.convolution_backward(
-                    dy, x, w, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+                    dy, x, wb, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
                     [False, True, False],
                 )[1]
             if dw.dtype != w.dtype:  # bf16 weights (shadow mode) / fp32 master
