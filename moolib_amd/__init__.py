@@ -356,6 +356,11 @@ class EnvPool(_EnvPoolCore):
             )
             self._dispatcher.start()
 
+    def __del__(self):
+        q = getattr(self, "_dispatch_q", None)
+        if q is not None:
+            q.put(None)  # unblock and end the dispatcher thread
+
     def _dispatch_loop(self):
         while True:
             item = self._dispatch_q.get()

@@ -81,11 +81,21 @@ def _identity(t):
     return ent
 
 
-def _materialize(pid, fd, nbytes, dtype, shape, strides, offset_el):
+def _materialize(pid, fd, nbytes, inode, dtype, shape, strides, offset_el):
     if pid == os.getpid():
         path = "/proc/self/fd/%d" % fd
     else:
         path = "/proc/%d/fd/%d" % (pid, fd)
+    # Guard against fd-number recycling: if the producer dropped the
+    # segment and the fd number now names something else, the inode (and
+    # usually size) won't match — fail loudly instead of aliasing
+    # unrelated memory.
+    st = os.stat(path)
+    if st.st_ino != inode or st.st_size < nbytes:
+        raise RuntimeError(
+            "memfd tensor expired: fd %d of pid %d no longer names the "
+            "shared segment (inode %d != %d)" % (fd, pid, st.st_ino, inode)
+        )
     esize = torch.empty(0, dtype=dtype).element_size()
     flat = torch.from_file(path, shared=True, size=nbytes // esize, dtype=dtype)
     return flat.as_strided(shape, strides, offset_el)
@@ -107,6 +117,7 @@ class SharedMemfdTensor:
                 os.getpid(),
                 self._fd,
                 self._nbytes,
+                os.fstat(self._fd).st_ino,
                 t.dtype,
                 tuple(t.shape),
                 tuple(t.stride()),
