@@ -290,7 +290,7 @@ __global__ void frames_u8_to_bf16_nhwc_kernel(const uint8_t* __restrict__ in,
 // One thread per output pixel computes all 16 output channels; the 1.2 KB
 // weight block is staged in LDS.
 
-__global__ void conv1_u8_nhwc_kernel(const uint8_t* __restrict__ in,   // [N,4,H,W]
+__global__ __launch_bounds__(256) void conv1_u8_nhwc_kernel(const uint8_t* __restrict__ in,   // [N,4,H,W]
                                      const hip_bfloat16* __restrict__ w,  // [3,3,4,16]
                                      const hip_bfloat16* __restrict__ bias,  // [16]
                                      hip_bfloat16* __restrict__ out,  // [N,H,W,16] (NHWC)
