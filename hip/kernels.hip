@@ -265,19 +265,23 @@ __global__ void maxpool3x3s2_bwd_kernel(const T* __restrict__ gout,
 
 __global__ void frames_u8_to_bf16_nhwc_kernel(const uint8_t* __restrict__ in,
                                               hip_bfloat16* __restrict__ out, float scale,
-                                              int N, int C, int H, int W) {
+                                              int N, int C, int H, int W, int CO) {
+  // CO >= C: output channels beyond C are zero (channel padding so the
+  // MFMA conv template, whose A fragments span 8 contiguous channels of
+  // one tap, can run the 4-channel first layer as C=8).
   int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t total = (int64_t)N * C * H * W;
+  int64_t total = (int64_t)N * CO * H * W;
   if (tid >= total) return;
   // tid enumerates the OUTPUT (NHWC) layout for coalesced writes.
-  int c = tid % C;
-  int64_t t = tid / C;
+  int c = tid % CO;
+  int64_t t = tid / CO;
   int w = t % W;
   t /= W;
   int h = t % H;
   int n = t / H;
-  uint8_t v = in[(((int64_t)n * C + c) * H + h) * W + w];
-  out[tid] = (hip_bfloat16)((float)v * scale);
+  float v = 0.f;
+  if (c < C) v = (float)in[(((int64_t)n * C + c) * H + h) * W + w] * scale;
+  out[tid] = (hip_bfloat16)v;
 }
 
 // ------------------------------------------ fused first conv (actor path)
@@ -479,20 +483,22 @@ void register_host_memory(at::Tensor t) {
   TORCH_CHECK(err == hipSuccess, "hipHostRegister failed: ", hipGetErrorString(err));
 }
 
-at::Tensor frames_u8_to_bf16_nhwc(at::Tensor x, double scale) {
+at::Tensor frames_u8_to_bf16_nhwc(at::Tensor x, double scale, int64_t padChannels) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4 && x.dtype() == at::kByte,
               "frames: 4D uint8 CUDA tensor expected");
   auto xc = x.contiguous();
   int N = xc.size(0), C = xc.size(1), H = xc.size(2), W = xc.size(3);
-  auto out = at::empty({N, C, H, W},
+  const int CO = padChannels > C ? (int)padChannels : C;
+  auto out = at::empty({N, CO, H, W},
                        xc.options().dtype(at::kBFloat16).memory_format(at::MemoryFormat::ChannelsLast));
-  int64_t total = (int64_t)N * C * H * W;
+  int64_t total = (int64_t)N * CO * H * W;
   int threads = 256;
   int64_t blocks = (total + threads - 1) / threads;
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   hipLaunchKernelGGL(frames_u8_to_bf16_nhwc_kernel, dim3(blocks), dim3(threads), 0, stream,
                      xc.data_ptr<uint8_t>(),
-                     reinterpret_cast<hip_bfloat16*>(out.data_ptr()), (float)scale, N, C, H, W);
+                     reinterpret_cast<hip_bfloat16*>(out.data_ptr()), (float)scale, N, C, H, W,
+                     CO);
   return out;
 }
 
@@ -893,10 +899,12 @@ using bf16 = __bf16;
 
 struct Desc {
   const bf16* w;  // [K, C, 3, 3], arbitrary strides (channels_last safe)
-  bf16* outF;     // fwd pack, ceil(9C/32)*32*K elems, or null
+  bf16* outF;     // fwd pack, ceil(9*Cp/32)*32*K elems, or null
   bf16* outD;     // dgrad pack, ceil(9K/32)*32*C elems, or null
   int64_t sK, sC, sH, sW;  // element strides of w
   int32_t C, K;
+  int32_t Cp;     // fwd pack channel count (>= C; channels C..Cp are zero —
+                  // the C=4 first conv packs as C=8 for the MFMA template)
 };
 constexpr int kMaxDescs = 20;
 struct Pack {
@@ -906,7 +914,7 @@ struct Pack {
 
 __global__ void repack3x3_kernel(Pack pack) {
   const Desc& d = pack.d[blockIdx.y];
-  const int KKf = (9 * d.C + 31) / 32;
+  const int KKf = (9 * d.Cp + 31) / 32;
   const int KKd = (9 * d.K + 31) / 32;
   const int nF = d.outF ? KKf * 32 * d.K : 0;
   const int nD = d.outD ? KKd * 32 * d.C : 0;
@@ -915,7 +923,7 @@ __global__ void repack3x3_kernel(Pack pack) {
     const bool isF = i < nF;
     const int idx = isF ? i : i - nF;
     const int KK = isF ? KKf : KKd;
-    const int CC = isF ? d.C : d.K;  // in-channels of this direction
+    const int CC = isF ? d.Cp : d.K;  // in-channels of this direction
     // idx = ((ct*KK + kk)*64 + lane)*8 + e
     const int e = idx & 7;
     const int lane = (idx >> 3) & 63;
@@ -929,7 +937,7 @@ __global__ void repack3x3_kernel(Pack pack) {
       const int tap = kd / CC, c = kd % CC;
       const int kh = tap / 3, kw = tap % 3;
       if (isF) {
-        v = d.w[n * d.sK + c * d.sC + kh * d.sH + kw * d.sW];
+        if (c < d.C) v = d.w[n * d.sK + c * d.sC + kh * d.sH + kw * d.sW];
       } else {
         // dgrad: W'[c_out=n... ] = w[c][n][2-kh][2-kw]
         v = d.w[c * d.sK + n * d.sC + (2 - kh) * d.sH + (2 - kw) * d.sW];
@@ -944,8 +952,9 @@ __global__ void repack3x3_kernel(Pack pack) {
 // Refresh fwd/dgrad MFMA packs for a batch of convs in ONE launch.
 // dgrad entries may be empty tensors (conv1 has no dgrad pack).
 void repack3x3_batched(std::vector<at::Tensor> ws, std::vector<at::Tensor> fwdBufs,
-                       std::vector<at::Tensor> dgradBufs) {
-  TORCH_CHECK(ws.size() == fwdBufs.size() && ws.size() == dgradBufs.size(),
+                       std::vector<at::Tensor> dgradBufs, std::vector<int64_t> packChannels) {
+  TORCH_CHECK(ws.size() == fwdBufs.size() && ws.size() == dgradBufs.size() &&
+                  ws.size() == packChannels.size(),
               "repack3x3: list length mismatch");
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   repack3x3::Pack pack;
@@ -969,10 +978,11 @@ void repack3x3_batched(std::vector<at::Tensor> ws, std::vector<at::Tensor> fwdBu
     e.sW = w.stride(3);
     e.C = C;
     e.K = K;
+    e.Cp = packChannels[i] > C ? (int)packChannels[i] : C;
     e.outF = nullptr;
     e.outD = nullptr;
     if (fwdBufs[i].defined() && fwdBufs[i].numel() > 0) {
-      TORCH_CHECK(fwdBufs[i].numel() == (9 * C + 31) / 32 * 32 * K, "repack3x3: fwd size");
+      TORCH_CHECK(fwdBufs[i].numel() == (9 * e.Cp + 31) / 32 * 32 * K, "repack3x3: fwd size");
       e.outF = reinterpret_cast<repack3x3::bf16*>(fwdBufs[i].data_ptr());
     }
     if (dgradBufs[i].defined() && dgradBufs[i].numel() > 0) {
@@ -997,7 +1007,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_relu_fwd", &bias_relu_fwd, "fused conv-bias + relu, one NHWC pass (gfx950)");
   m.def("bias_relu_bwd", &bias_relu_bwd, "bias_relu backward: dx + fp32 db in one pass");
   m.def("bias_add2_fwd", &bias_add2_fwd, "fused residual close: x + bias1 + shortcut (+ bias2)");
-  m.def("frames_u8_to_bf16_nhwc", &frames_u8_to_bf16_nhwc, "fused uint8->bf16 NHWC scale");
+  m.def("frames_u8_to_bf16_nhwc", &frames_u8_to_bf16_nhwc,
+        "fused uint8->bf16 NHWC scale (optionally zero-padding channels)",
+        py::arg("x"), py::arg("scale"), py::arg("pad_channels") = 0);
   m.def("conv1_u8_nhwc", &conv1_u8_nhwc, "fused uint8 frames -> conv(4->16,3x3)+bias, NHWC bf16");
   m.def("conv3x3_nhwc_fused", &conv3x3_nhwc_fused,
         "MFMA 3x3/s1/p1 NHWC conv with fused relu/bias prologue + bias/relu/residual epilogue",

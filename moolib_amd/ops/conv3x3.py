@@ -93,36 +93,52 @@ def repack(module):
     cat per conv, ~14 kernels each) remains the fallback."""
     import torch.nn as nn
 
-    ws, fs, ds, caches = [], [], [], []
+    ws, fs, ds, cps, caches = [], [], [], [], []
     fallback = []
     for m in module.modules():
         if isinstance(m, nn.Conv2d):
             cache = getattr(m, "_c3_cache", None)
             dcache = getattr(m, "_c3_dgrad_cache", None)
-            if cache is None and dcache is None:
+            c1cache = getattr(m, "_c1_pack_cache", None)
+            if cache is None and dcache is None and c1cache is None:
                 continue
             w = m.weight.detach()
             if w.is_cuda and w.dtype == torch.bfloat16:
-                ws.append(w)
-                fs.append(cache[1] if cache is not None else w.new_empty(0))
-                ds.append(dcache[1] if dcache is not None else w.new_empty(0))
-                caches.append((m, cache, dcache))
+                if cache is not None or dcache is not None:
+                    ws.append(w)
+                    fs.append(cache[1] if cache is not None else w.new_empty(0))
+                    ds.append(dcache[1] if dcache is not None else w.new_empty(0))
+                    cps.append(0)
+                    caches.append((m, cache, dcache))
+                if c1cache is not None:
+                    # the C=8-padded first-conv pack rides the same launch
+                    ws.append(w)
+                    fs.append(c1cache[1])
+                    ds.append(w.new_empty(0))
+                    cps.append(8)
+                    caches.append((m, c1cache, None))
             else:
-                fallback.append((m, cache, dcache))
+                fallback.append((m, cache, dcache, c1cache))
     if ws:
-        _k().repack3x3_batched(ws, fs, ds)
+        _k().repack3x3_batched(ws, fs, ds, cps)
         for m, cache, dcache in caches:
             if cache is not None:
                 cache[0] = m.weight._version
             if dcache is not None:
                 dcache[0] = m.weight._version
-    for m, cache, dcache in fallback:
+    for m, cache, dcache, c1cache in fallback:
         if cache is not None:
             cache[1].copy_(pack_weight(m.weight.detach()))
             cache[0] = m.weight._version
         if dcache is not None:
             dcache[1].copy_(pack_weight_dgrad(m.weight.detach()))
             dcache[0] = m.weight._version
+        if c1cache is not None:
+            import torch.nn.functional as F
+
+            w8 = F.pad(m.weight.detach(), (0, 0, 0, 0, 0, 8 - m.weight.shape[1]))
+            c1cache[1].copy_(pack_weight(w8))
+            c1cache[0] = m.weight._version
 
 
 def pack_weight_dgrad(w):
@@ -206,36 +222,66 @@ def conv3x3_autograd(x, conv_module):
     return _Conv3x3Fn.apply(x, conv_module.weight, conv_module)
 
 
+def conv1_packed_buffer(conv):
+    """C=8 zero-padded MFMA pack for the 4-channel first conv.
+
+    The MFMA template's A fragment spans 8 contiguous channels of one
+    tap, so C=4 runs as C=8 with zero channels (25% padded MACs beat the
+    scalar direct conv by a wide margin at learner batch). Version-checked
+    like packed_buffer; refreshed by repack() (the batched repack kernel
+    zero-fills channels >= the real C)."""
+    import torch.nn.functional as F
+
+    w = conv.weight
+    ver = w._version
+    cache = getattr(conv, "_c1_pack_cache", None)
+    if cache is None or cache[0] != ver:
+        w8 = F.pad(w.detach(), (0, 0, 0, 0, 0, 8 - w.shape[1]))
+        packed = pack_weight(w8)
+        if cache is None:
+            conv._c1_pack_cache = cache = [ver, packed]
+        else:
+            cache[1].copy_(packed)
+            cache[0] = ver
+    return cache[1]
+
+
 class _Conv1U8Fn(torch.autograd.Function):
-    """First conv (uint8 frames -> 16ch) with fwd on the fused conv1_u8
-    kernel and dW/db on our wgrad kernel — no MIOpen anywhere in the layer
-    (frames never need an input gradient)."""
+    """First conv (uint8 frames -> 16ch), learner path: frames decode to a
+    C=8 zero-padded bf16 NHWC tensor (one fused kernel) and the conv runs
+    on the MFMA template with the bias fused into the epilogue; dW comes
+    from our wgrad kernel on the saved padded activation, db is a sum.
+    No MIOpen anywhere in the layer (frames never need an input
+    gradient). The scalar conv1_u8 kernel remains the actor (no-grad,
+    hipGraph-captured) path where N is small."""
 
     @staticmethod
-    def forward(ctx, x_u8, weight, bias, scale):
+    def forward(ctx, x_u8, weight, bias, conv_module, scale):
         k = _k()
-        w = weight.detach().to(torch.bfloat16).permute(2, 3, 1, 0).contiguous()
-        b = bias.detach().to(torch.bfloat16)
-        y = k.conv1_u8_nhwc(x_u8, w, b, scale)
-        ctx.save_for_backward(x_u8, weight)
-        ctx.scale = scale
+        xb8 = k.frames_u8_to_bf16_nhwc(x_u8, scale, 8)
+        y = conv3x3(
+            xb8, conv1_packed_buffer(conv_module), weight.shape[0],
+            epi=EPI_BIAS, bias1=bias.detach(),
+        )
+        ctx.save_for_backward(xb8, weight)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x_u8, w = ctx.saved_tensors
+        xb8, w = ctx.saved_tensors
         k = _k()
         dyb = dy.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
-        xb = k.frames_u8_to_bf16_nhwc(x_u8, ctx.scale)  # same scaling as fwd
         K, C = w.shape[0], w.shape[1]
-        g = k.wgrad3x3_nhwc(xb, dyb)[: 9 * C]
-        dw = g.view(3, 3, C, K).permute(3, 2, 0, 1).to(w.dtype)
+        g = k.wgrad3x3_nhwc(xb8, dyb)[: 9 * 8]
+        dw = g.view(3, 3, 8, K)[:, :, :C].permute(3, 2, 0, 1).to(w.dtype)
         db = dy.float().sum((0, 2, 3)).to(w.dtype)
-        return None, dw, db, None
+        return None, dw, db, None, None
 
 
 def conv1_u8_autograd(x_u8, conv_module, scale):
-    return _Conv1U8Fn.apply(x_u8, conv_module.weight, conv_module.bias, scale)
+    return _Conv1U8Fn.apply(
+        x_u8, conv_module.weight, conv_module.bias, conv_module, scale
+    )
 
 
 def conv3x3(x, w_packed, k, relu_in=False, bias_in=None, epi=EPI_NONE,

@@ -877,12 +877,27 @@ class TestBatchedRepack:
             fs.append(torch.empty_like(f_ref))
             ds.append(torch.empty_like(d_ref) if d_ref is not None else w.new_empty(0))
             refs.append((f_ref, d_ref))
-        _kernels.repack3x3_batched(ws, fs, ds)
+        _kernels.repack3x3_batched(ws, fs, ds, [0] * len(ws))
         torch.cuda.synchronize()
         for (C, K), fbuf, dbuf, (f_ref, d_ref) in zip(specs, fs, ds, refs):
             assert torch.equal(fbuf, f_ref), (C, K, "fwd")
             if d_ref is not None:
                 assert torch.equal(dbuf, d_ref), (C, K, "dgrad")
+
+    def test_padded_first_conv_pack(self):
+        """pack_channels=8 for a C=4 weight must equal packing the
+        zero-padded weight (the conv1 C=8 path)."""
+        import torch.nn.functional as F
+
+        from moolib_amd import _kernels
+        from moolib_amd.ops import conv3x3 as c3
+
+        w = torch.randn(16, 4, 3, 3, device="cuda", dtype=torch.bfloat16)
+        ref = c3.pack_weight(F.pad(w, (0, 0, 0, 0, 0, 4)))
+        buf = torch.empty_like(ref)
+        _kernels.repack3x3_batched([w], [buf], [w.new_empty(0)], [8])
+        torch.cuda.synchronize()
+        assert torch.equal(buf, ref)
 
 
 @gpu
