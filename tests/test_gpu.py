@@ -939,3 +939,17 @@ print("OK")
         r = subprocess.run([sys.executable, "-c", code], capture_output=True, text=True,
                            timeout=240, env=env, cwd=repo)
         assert r.returncode == 0 and "OK" in r.stdout, r.stderr[-2000:]
+
+
+@gpu
+@requires_gpu
+class TestFramesPad8:
+    def test_matches_generic(self):
+        from moolib_amd import _kernels
+
+        x = torch.randint(0, 256, (5, 4, 84, 84), dtype=torch.uint8, device="cuda")
+        got = _kernels.frames_u8_to_bf16_nhwc(x, 1.0 / 255.0, 8)
+        want4 = (x.float() / 255.0).bfloat16()
+        assert got.shape == (5, 8, 84, 84)
+        assert torch.equal(got[:, :4].float().cpu(), want4.float().cpu())
+        assert got[:, 4:].abs().sum().item() == 0.0
