@@ -184,14 +184,24 @@ class AtariNet(nn.Module):
             # touches MIOpen in either direction.
             sec0 = self.sections[0]
             c1 = sec0.conv
-            if torch.is_grad_enabled() and c1.weight.requires_grad:
-                from moolib_amd.ops import conv3x3 as c3m
+            from moolib_amd.ops import conv3x3 as c3m
 
+            if torch.is_grad_enabled() and c1.weight.requires_grad:
                 x = c3m.conv1_u8_autograd(x, c1, 1.0 / 255.0)
-            else:
+            elif _os.environ.get("MOOLIB_AMD_CONV1_SCALAR"):
+                # the original fused scalar kernel (kept for A/Bs)
                 w = c1.weight.detach().to(torch.bfloat16).permute(2, 3, 1, 0).contiguous()
                 b = c1.bias.detach().to(torch.bfloat16)
                 x = kernels.conv1_u8_nhwc(x, w, b, 1.0 / 255.0)
+            else:
+                # same C=8 MFMA route as the learner (graph-capturable:
+                # the packed weights live in a stable version-checked
+                # buffer refreshed by repack())
+                x8 = kernels.frames_u8_to_bf16_nhwc(x, 1.0 / 255.0, 8)
+                x = c3m.conv3x3(
+                    x8, c3m.conv1_packed_buffer(c1), c1.out_channels,
+                    epi=c3m.EPI_BIAS, bias1=c1.bias.detach(),
+                )
             from moolib_amd.ops.pool import maxpool3x3s2
 
             x = maxpool3x3s2(x)
