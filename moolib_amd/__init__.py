@@ -49,7 +49,16 @@ from .utils import nest
 # Group / AllReduce --------------------------------------------------------
 
 Group = _core.Group
-AllReduce = Future  # group.all_reduce returns a Future-shaped handle
+
+
+class AllReduce(Future):
+    """Handle for an in-flight collective (reference src/group.h:493-499).
+
+    `group.all_reduce(name, value, op)` returns this Future: `result()`
+    blocks for the reduced value, `done()/exception()/await` as usual; a
+    membership change cancels in-flight reductions with an error (retry
+    under the new sync_id). Subclass of Future so isinstance checks work
+    both ways."""
 
 
 # Future / Queue awaitability ---------------------------------------------
