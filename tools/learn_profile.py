@@ -53,6 +53,16 @@ def main():
     print(prof.key_averages(group_by_input_shape=False).table(
         sort_by="self_cuda_time_total", row_limit=35))
 
+    # second pass WITH CPU activity: attributes CUDA time to aten ops (and
+    # their shapes), which names the generic elementwise kernels
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                 record_shapes=True) as prof2:
+        for _ in range(3):
+            peer.compute_gradients(data)
+        torch.cuda.synchronize()
+    print(prof2.key_averages(group_by_input_shape=True).table(
+        sort_by="cuda_time_total", row_limit=45))
+
 
 if __name__ == "__main__":
     main()
