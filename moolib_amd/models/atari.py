@@ -276,15 +276,21 @@ class AtariNet(nn.Module):
 
         policy_logits = self.policy(core_output).float()
         baseline = self.baseline(core_output).float()
-        action = torch.multinomial(
-            F.softmax(policy_logits, dim=-1), num_samples=1, generator=self.sample_generator
-        )
-
         out = dict(
             policy_logits=policy_logits.view(T, B, self.num_actions),
             baseline=baseline.view(T, B),
-            action=action.view(T, B),
         )
+        if not torch.is_grad_enabled():
+            # Actor path: sample the action in-forward (reference
+            # models.py:134-140). The learner discards the sample (it
+            # trains on the ACTOR's actions), and multinomial is one of
+            # the slowest ops in the step (~230 us CPU+GPU at [672, 18]) —
+            # skip it whenever autograd is recording.
+            action = torch.multinomial(
+                F.softmax(policy_logits, dim=-1), num_samples=1,
+                generator=self.sample_generator,
+            )
+            out["action"] = action.view(T, B)
         return out, core_state
 
 
