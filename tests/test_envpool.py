@@ -3,6 +3,8 @@
 Mirrors the reference's test/unit/test_envpool.py strategy: validate
 double-buffered stepping semantics, auto-reset behavior, and field shapes.
 """
+import time
+
 import numpy as np
 import pytest
 import torch
@@ -270,3 +272,20 @@ class TestExternalEnvRunner:
         assert obs["state"].shape == (2, 2)
         assert r1.running()
         del pool
+
+
+class TestPoll:
+    def test_poll_semantics(self):
+        """poll() is a non-consuming completion check: false before the
+        step completes/after result() consumed it, true in between."""
+        pool = moolib_amd.EnvPool(CountingEnv, num_processes=2, batch_size=4, num_batches=1)
+        assert not pool.poll(0)  # nothing stepped yet
+        fut = pool.step(0, torch.zeros(4, dtype=torch.int64))
+        t0 = time.time()
+        while not pool.poll(0) and time.time() - t0 < 30:
+            time.sleep(0.002)
+        assert pool.poll(0)      # complete, not yet consumed
+        assert pool.poll(0)      # non-consuming: still true
+        fut.result()
+        assert not pool.poll(0)  # consumed
+        assert not pool.poll(7)  # out of range
