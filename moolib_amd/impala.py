@@ -83,9 +83,9 @@ class ImpalaConfig:
     #   re-enable via graph_learner=True to investigate (round 2).
     pinned_staging: bool = True  # pinned bounce buffers for shm->HBM copies
     shm_host_register: bool = False  # hipHostRegister the env shm (measured SLOWER on MI355X)
-    prewarm: bool = True  # construction-time dry runs: settle MIOpen find /
-    #   graph captures / allocator before the first real step (weights and
-    #   optimizer state are restored in place afterwards)
+    prewarm: bool = True  # construction-time pipeline priming: act until the
+    #   learn backlog is at its steady level, so measurement windows start
+    #   at steady state instead of spending ~30 steps filling it
     actor_side_stream: bool = False  # overlap actor work on a side HIP stream.
     #   OFF by default: concurrent hipGraph replays on two streams still hit
     #   an intermittent HSA memory fault (~1 in 8 runs) on ROCm 7.0 even
@@ -351,80 +351,25 @@ class ImpalaPeer:
                 # observation shape differs from the synthetic warm batch
                 logging.warning("prewarm skipped: %s", e)
 
-    def _prewarm(self, iters=15):
-        """One-time warm at construction: MIOpen's exhaustive find (which
-        trickles across the first ~10 learner steps otherwise), hipGraph
-        captures, allocator pools and weight packs all settle on synthetic
-        data; weights/optimizer state are then restored IN PLACE (same
-        tensor addresses, so captured graphs stay valid) and stats reset.
-        Measured: without this, a 20-step window after 5 warmup steps reads
-        ~79k f/s while true steady state is ~93k (same box,
-        profiles/evidence/r3r_*.json)."""
-        cfg = self.cfg
-        dev = cfg.device
-        saved_params = [p.detach().clone() for p in self.model.parameters()]
-        saved_bufs = [b.detach().clone() for b in self.model.buffers()]
-        T, B, A = cfg.unroll_length, cfg.batch_size, cfg.num_actions
-        dt = next(self.fwd_model.parameters()).dtype
-        data = {
-            "env_outputs": {
-                "state": torch.randint(0, 255, (T + 1, B, 4, 84, 84),
-                                       dtype=torch.uint8, device=dev),
-                "reward": torch.zeros(T + 1, B, device=dev),
-                "done": torch.zeros(T + 1, B, dtype=torch.bool, device=dev),
-                "prev_action": torch.zeros(T + 1, B, dtype=torch.int64, device=dev),
-            },
-            "actor_outputs": {
-                "policy_logits": torch.randn(T + 1, B, A, device=dev),
-                "action": torch.zeros(T + 1, B, dtype=torch.int64, device=dev),
-            },
-            "initial_core_state": tuple(
-                s.to(device=dev, dtype=dt)
-                for s in self.fwd_model.initial_state(batch_size=B)
-            ),
-        }
-        AB = cfg.actor_batch_size
-        act_env = {
-            "state": torch.randint(0, 255, (AB, 4, 84, 84), dtype=torch.uint8, device=dev),
-            "reward": torch.zeros(AB, device=dev),
-            "done": torch.zeros(AB, dtype=torch.bool, device=dev),
-            "prev_action": torch.zeros(AB, dtype=torch.int64, device=dev),
-        }
-        act_core = tuple(
-            s.to(device=dev, dtype=dt)
-            for s in self.fwd_model.initial_state(batch_size=AB)
-        )
-        for _ in range(iters):
-            self._actor_call({"env": act_env, "core": act_core})
-            self.compute_gradients(data)
-            self._opt_call({})
-        torch.cuda.synchronize()
-        with torch.no_grad():
-            for p, s in zip(self.model.parameters(), saved_params):
-                p.data.copy_(s)
-            for b, s in zip(self.model.buffers(), saved_bufs):
-                b.copy_(s)
-            # reset optimizer state in place (addresses survive for the
-            # captured optimizer graph; zeroed step restarts bias correction)
-            for group in self.optimizer.param_groups:
-                for p in group["params"]:
-                    st = self.optimizer.state.get(p)
-                    if st:
-                        for v in st.values():
-                            if torch.is_tensor(v):
-                                v.zero_()
-            for p in self.model.parameters():
-                if p.grad is not None:
-                    p.grad.zero_()
-            if self.bf16_shadow:
-                for p in self._fwd_params:
-                    if p.grad is not None:
-                        p.grad.zero_()
-            self._sync_shadow()
-            from moolib_amd.ops import conv3x3 as _c3
+    def _prewarm(self):
+        """Fill the actor->learner pipeline to its steady backlog at
+        construction.
 
-            _c3.repack(self.fwd_model)
-        self.stats = make_stats()
+        Measured (profiles/evidence/r3t_*.json, one box): a 20-step window
+        after 5 warmup steps reads ~80k f/s while the same window after 20
+        warmup steps reads 92-95k. The gap is NOT one-time init (a dry
+        fwd/bwd/opt prewarm changed nothing — MIOpen find and the graph
+        captures already settle inside 5 warmup steps); it is the learn
+        backlog filling: the actor gets ahead of the learner only by its
+        small per-step surplus, so reaching the steady max_learn_backlog
+        takes ~30-40 optimizer steps. Acting the pipeline full at
+        construction starts ANY window at steady state — no borrowed work,
+        since in steady state the backlog is level across the window (the
+        actor replaces every batch the learner consumes)."""
+        target = self.cfg.max_learn_backlog
+        t0 = time.time()
+        while self.learn_batcher.size() < target and time.time() - t0 < 120.0:
+            self.act_once()
         torch.cuda.synchronize()
 
     def _t(self, name, t0):
