@@ -561,10 +561,14 @@ class ImpalaPeer:
         # stalled 24 ms in the burst — profiles/evidence/r3v_ramp.txt).
         # The act's GPU work is small and queues behind the learner's on
         # the same stream; its CPU orchestration overlaps learner GPU time.
-        if (
-            self.learn_batcher.size() < cfg.max_learn_backlog
-            and self._env_ready()
-        ):
+        # (~5 acts sustain one optimizer step; a single act per iteration
+        # still left ~35-act bursts at the drain point)
+        for _ in range(2 * cfg.num_actor_batches):
+            if (
+                self.learn_batcher.size() >= cfg.max_learn_backlog
+                or not self._env_ready()
+            ):
+                break
             self.act_once()
 
         t0 = time.perf_counter() if self.profile else 0.0

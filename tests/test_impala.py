@@ -41,14 +41,15 @@ def test_impala_cpu_end_to_end():
     )
 
     t0 = time.time()
-    events = {"optimize": 0, "learn": 0, "act": 0, "idle": 0}
+    events = {"optimize": 0, "learn": 0, "act": 0, "idle": 0, "throttle": 0}
     while events["optimize"] < 3 and time.time() - t0 < 200:
         ev = peer.step_once()
         events[ev] += 1
 
     assert events["optimize"] >= 3, events
     assert events["learn"] >= 3
-    assert events["act"] > 10
+    # acting may happen opportunistically inside learn/optimize iterations
+    # (env-fed scheduling), so count env frames rather than "act" returns
     assert peer.stats["env_train_steps"].result() >= 3 * cfg.unroll_length * cfg.batch_size
     assert peer.stats["optimizer_steps"].result() == events["optimize"]
     # parameters actually moved
