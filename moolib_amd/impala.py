@@ -343,6 +343,9 @@ class ImpalaPeer:
         # Phase profiling (bench --breakdown): cumulative seconds per phase,
         # with device sync at boundaries so GPU time is attributed correctly.
         self.profile = False
+        # env-fed scheduling (opportunistic acting before learn/optimize);
+        # MOOLIB_AMD_ENVFED=0 restores strict learn-priority for A/Bs
+        self._envfed = os.environ.get("MOOLIB_AMD_ENVFED", "1") != "0"
         self.phase_times = {}
         if self.is_cuda and cfg.prewarm and not os.environ.get("MOOLIB_AMD_NO_PREWARM"):
             try:
@@ -563,13 +566,14 @@ class ImpalaPeer:
         # the same stream; its CPU orchestration overlaps learner GPU time.
         # (~5 acts sustain one optimizer step; a single act per iteration
         # still left ~35-act bursts at the drain point)
-        for _ in range(2 * cfg.num_actor_batches):
-            if (
-                self.learn_batcher.size() >= cfg.max_learn_backlog
-                or not self._env_ready()
-            ):
-                break
-            self.act_once()
+        if self._envfed:
+            for _ in range(2 * cfg.num_actor_batches):
+                if (
+                    self.learn_batcher.size() >= cfg.max_learn_backlog
+                    or not self._env_ready()
+                ):
+                    break
+                self.act_once()
 
         t0 = time.perf_counter() if self.profile else 0.0
         if acc.has_gradients():
