@@ -343,9 +343,14 @@ class ImpalaPeer:
         # Phase profiling (bench --breakdown): cumulative seconds per phase,
         # with device sync at boundaries so GPU time is attributed correctly.
         self.profile = False
-        # env-fed scheduling (opportunistic acting before learn/optimize);
-        # MOOLIB_AMD_ENVFED=0 restores strict learn-priority for A/Bs
-        self._envfed = os.environ.get("MOOLIB_AMD_ENVFED", "1") != "0"
+        # Env-fed scheduling (opportunistic acting before learn/optimize)
+        # smooths the periodic act bursts but LOSES throughput overall —
+        # same-box interleaved A/B: off 81.1/82.9k short, 88.5k sustained
+        # vs on 79.5/71.1k short, 84.9k sustained (evidence/r3y_*.json).
+        # The inline acts' CPU orchestration serializes with the learner's
+        # and costs more than the bursts they spread out. Off by default;
+        # MOOLIB_AMD_ENVFED=1 enables it for experiments.
+        self._envfed = os.environ.get("MOOLIB_AMD_ENVFED", "0") == "1"
         self.phase_times = {}
         if self.is_cuda and cfg.prewarm and not os.environ.get("MOOLIB_AMD_NO_PREWARM"):
             try:
