@@ -161,6 +161,44 @@ def make_stats():
     }
 
 
+def flatten_master_shadow(master_params, fwd_params):
+    """Re-home master (fp32) and shadow (bf16) params and their grads as
+    views into four contiguous flat buffers.
+
+    Cross-dtype ``_foreach_copy_`` decomposes into one copy kernel per
+    tensor, so the per-optimizer-step fp32->bf16 weight sync and
+    bf16->fp32 grad cast each cost ~1 launch per parameter (measured 38
+    aten::copy_ calls/step, profiles/evidence/r4d_stepprof.txt). With
+    flat buffers each cast is ONE kernel: ``flat_shadow.copy_(flat_master)``
+    and ``flat_grad32.copy_(flat_grad16)``. Grad views are preassigned so
+    autograd accumulates straight into the flat buffer.
+
+    Returns (flat_master, flat_shadow, flat_grad32, flat_grad16).
+    Note: views are contiguous — a channels_last weight loses its NHWC
+    strides, which is fine here because every forward path reads the
+    MFMA-packed weight caches (ops/conv3x3), not the raw strides.
+    """
+    total = sum(p.numel() for p in fwd_params)
+    dev = fwd_params[0].device if fwd_params else "cpu"
+    flat_master = torch.empty(total, dtype=torch.float32, device=dev)
+    flat_shadow = torch.empty(total, dtype=torch.bfloat16, device=dev)
+    flat_grad32 = torch.zeros(total, dtype=torch.float32, device=dev)
+    flat_grad16 = torch.zeros(total, dtype=torch.bfloat16, device=dev)
+    off = 0
+    for pm, pf in zip(master_params, fwd_params):
+        n = pm.numel()
+        mv = flat_master[off : off + n].view_as(pm)
+        mv.copy_(pm.detach())
+        pm.data = mv
+        pm.grad = flat_grad32[off : off + n].view_as(pm)
+        sv = flat_shadow[off : off + n].view_as(pf)
+        sv.copy_(pf.detach())
+        pf.data = sv
+        pf.grad = flat_grad16[off : off + n].view_as(pf)
+        off += n
+    return flat_master, flat_shadow, flat_grad32, flat_grad16
+
+
 class ImpalaPeer:
     """One learner+actor peer (one GPU)."""
 
@@ -281,24 +319,12 @@ class ImpalaPeer:
             # of one per tensor (cross-dtype _foreach_copy_ decomposes to
             # per-tensor copies — measured 38 copy launches per optimizer
             # step in profiles/evidence/r4d_stepprof.txt).
-            total = sum(p.numel() for p in self._fwd_params)
-            dev = cfg.device
-            self._flat_master = torch.empty(total, dtype=torch.float32, device=dev)
-            self._flat_shadow = torch.empty(total, dtype=torch.bfloat16, device=dev)
-            self._flat_grad32 = torch.zeros(total, dtype=torch.float32, device=dev)
-            self._flat_grad16 = torch.zeros(total, dtype=torch.bfloat16, device=dev)
-            off = 0
-            for pm, pf in zip(self._master_params, self._fwd_params):
-                n = pm.numel()
-                mv = self._flat_master[off : off + n].view_as(pm)
-                mv.copy_(pm.detach())
-                pm.data = mv
-                pm.grad = self._flat_grad32[off : off + n].view_as(pm)
-                sv = self._flat_shadow[off : off + n].view_as(pf)
-                sv.copy_(pf.detach())
-                pf.data = sv
-                pf.grad = self._flat_grad16[off : off + n].view_as(pf)
-                off += n
+            (
+                self._flat_master,
+                self._flat_shadow,
+                self._flat_grad32,
+                self._flat_grad16,
+            ) = flatten_master_shadow(self._master_params, self._fwd_params)
         else:
             self.fwd_model = self.model
             self._master_params = []

@@ -56,3 +56,58 @@ def test_impala_cpu_end_to_end():
     total_norm = sum(p.detach().norm().item() for p in peer.model.parameters())
     assert total_norm == total_norm  # not NaN
     assert peer.model_version == events["optimize"]
+
+def test_flatten_master_shadow_cpu():
+    """Flat-buffer shadow layout (bf16 learner precision mode): params keep
+    their values, master<->shadow casts are single flat copies, and
+    autograd accumulates into the preassigned flat grad views."""
+    import copy
+
+    import torch
+
+    from moolib_amd.impala import flatten_master_shadow
+
+    torch.manual_seed(11)
+    master = torch.nn.Sequential(
+        torch.nn.Linear(8, 16), torch.nn.ReLU(), torch.nn.Linear(16, 4)
+    )
+    shadow = copy.deepcopy(master).to(torch.bfloat16)
+    mp = [p for p in master.parameters() if p.requires_grad]
+    sp = [p for p in shadow.parameters() if p.requires_grad]
+    before = [p.detach().clone() for p in mp]
+
+    fm, fs, g32, g16 = flatten_master_shadow(mp, sp)
+
+    # values preserved; storage re-homed into the flat buffers
+    for p, b in zip(mp, before):
+        assert torch.equal(p.detach(), b)
+        assert p.data_ptr() >= fm.data_ptr()
+        assert p.grad is not None and p.grad.shape == p.shape
+    for p in sp:
+        assert p.data_ptr() >= fs.data_ptr()
+
+    # optimizer step on master params mutates the flat buffer in place
+    opt = torch.optim.SGD(mp, lr=0.1)
+    x = torch.randn(3, 8)
+    master(x).sum().backward()
+    assert g32.abs().sum() > 0  # autograd landed in the flat fp32 buffer
+    opt.step()
+
+    # one-shot sync: flat cast copy == per-param fp32->bf16 cast
+    fs.copy_(fm)
+    for pm, pf in zip(mp, sp):
+        assert torch.equal(pf.detach(), pm.detach().to(torch.bfloat16))
+
+    # shadow backward accumulates into the flat bf16 buffer; zeroing the
+    # flat buffer zeroes every grad view
+    shadow(x.to(torch.bfloat16)).float().sum().backward()
+    assert g16.abs().sum() > 0
+    assert all(p.grad.abs().sum() > 0 for p in sp)
+    g16.zero_()
+    assert all(p.grad.abs().sum() == 0 for p in sp)
+
+    # grad cast path: flat bf16 -> flat fp32 equals per-param casts
+    shadow(x.to(torch.bfloat16)).float().sum().backward()
+    g32.copy_(g16)
+    for pm, pf in zip(mp, sp):
+        assert torch.equal(pm.grad, pf.grad.float())
