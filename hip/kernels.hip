@@ -719,8 +719,13 @@ std::vector<at::Tensor> bias_relu_bwd(at::Tensor dy, at::Tensor y) {
   int64_t total8 = y.numel() / 8;
   int threads = 256;
   // grid-stride: enough workgroups to fill 256 CUs across 8 XCDs without
-  // letting the per-block channel atomics dominate
-  int64_t blocks = std::min<int64_t>((total8 + threads - 1) / threads, 2048);
+  // letting the per-block channel atomics dominate (cap swept via
+  // MOOLIB_AMD_BIAS_BWD_BLOCKS; 2048 measured best, tools/bias_micro.py)
+  static const int64_t blockCap = []() {
+    const char* s = std::getenv("MOOLIB_AMD_BIAS_BWD_BLOCKS");
+    return s ? std::strtoll(s, nullptr, 10) : (int64_t)2048;
+  }();
+  int64_t blocks = std::min<int64_t>((total8 + threads - 1) / threads, blockCap);
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
       at::kBFloat16, at::kHalf, y.scalar_type(), "bias_relu_bwd", [&] {
