@@ -718,13 +718,19 @@ std::vector<at::Tensor> bias_relu_bwd(at::Tensor dy, at::Tensor y) {
   auto db = at::zeros({C}, y.options().dtype(at::kFloat));
   int64_t total8 = y.numel() / 8;
   int threads = 256;
-  // grid-stride: enough workgroups to fill 256 CUs across 8 XCDs without
-  // letting the per-block channel atomics dominate (cap swept via
-  // MOOLIB_AMD_BIAS_BWD_BLOCKS; 2048 measured best, tools/bias_micro.py)
-  static const int64_t blockCap = []() {
+  // grid-stride with a SMALL grid: each block ends with one global atomic
+  // per channel, so the per-block finalization atomics — not occupancy —
+  // set the floor. Swept caps 256..8192 at the three learner shapes
+  // (profiles/evidence/r4{i,j}_bias_micro.txt): per-shape optima
+  // 768/384/256 fit blocks = clamp(total8/3000, 256, 768), 29-51% faster
+  // than the old 2048 cap. MOOLIB_AMD_BIAS_BWD_BLOCKS overrides for A/Bs.
+  static const int64_t blockCapEnv = []() {
     const char* s = std::getenv("MOOLIB_AMD_BIAS_BWD_BLOCKS");
-    return s ? std::strtoll(s, nullptr, 10) : (int64_t)2048;
+    return s ? std::strtoll(s, nullptr, 10) : (int64_t)0;
   }();
+  int64_t blockCap = blockCapEnv > 0
+                         ? blockCapEnv
+                         : std::min<int64_t>(std::max<int64_t>(total8 / 3000, 256), 768);
   int64_t blocks = std::min<int64_t>((total8 + threads - 1) / threads, blockCap);
   hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
   AT_DISPATCH_FLOATING_TYPES_AND2(
