@@ -658,6 +658,32 @@ __global__ void bias_relu_bwd_kernel(const T* __restrict__ dy, const T* __restri
   if (threadIdx.x < C) atomicAdd(&db[threadIdx.x], lds_db[threadIdx.x]);
 }
 
+// Per-channel sum of an NHWC tensor (bias gradients): same small-grid +
+// LDS + one-global-atomic-per-channel layout as bias_relu_bwd_kernel.
+// aten's bf16 reduce runs this shape at ~0.4 TB/s (22.5 us at
+// [672,32,21,21], profiles/evidence/r3i_learnprof.txt).
+template <typename T>
+__global__ void channel_sum_kernel(const T* __restrict__ x, float* __restrict__ out,
+                                   int64_t total8, int c8n) {
+  __shared__ float lds[64];
+  int C = c8n * 8;
+  if (threadIdx.x < C) lds[threadIdx.x] = 0.f;
+  __syncthreads();
+  int64_t stride = (int64_t)blockDim.x * gridDim.x;  // multiple of c8n
+  int64_t tid0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  int c8 = (int)(tid0 % c8n);
+  for (int64_t tid = tid0; tid < total8; tid += stride) {
+    const Vec8<T> v = *reinterpret_cast<const Vec8<T>*>(x + tid * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += (float)v.v[j];
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(&lds[c8 * 8 + j], acc[j]);
+  __syncthreads();
+  if (threadIdx.x < C) atomicAdd(&out[threadIdx.x], lds[threadIdx.x]);
+}
+
 // y = x + b1 + s (+ b2): residual close with the producing conv's bias and,
 // when the shortcut came through a pooled bias-free section conv, that
 // conv's pending bias too.
@@ -740,6 +766,26 @@ std::vector<at::Tensor> bias_relu_bwd(at::Tensor dy, at::Tensor y) {
                            dx.data_ptr<scalar_t>(), db.data_ptr<float>(), total8, C / 8);
       });
   return {dx, db};
+}
+
+at::Tensor channel_sum_fp32(at::Tensor x) {
+  checkNhwcPair(x, "channel_sum");
+  int C = x.size(1);
+  TORCH_CHECK(C <= 64, "channel_sum: C <= 64 (LDS accumulator)");
+  auto out = at::zeros({C}, x.options().dtype(at::kFloat));
+  int64_t total8 = x.numel() / 8;
+  int threads = 256;
+  // small grid: the per-block channel atomics set the floor (see
+  // bias_relu_bwd above; same sweep, profiles/evidence/r4j_bias_micro.txt)
+  int64_t blockCap = std::min<int64_t>(std::max<int64_t>(total8 / 3000, 256), 768);
+  int64_t blocks = std::min<int64_t>((total8 + threads - 1) / threads, blockCap);
+  hipStream_t stream = c10::hip::getCurrentHIPStream().stream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::kBFloat16, at::kHalf, x.scalar_type(), "channel_sum", [&] {
+        hipLaunchKernelGGL(channel_sum_kernel<scalar_t>, dim3(blocks), dim3(threads), 0, stream,
+                           x.data_ptr<scalar_t>(), out.data_ptr<float>(), total8, C / 8);
+      });
+  return out;
 }
 
 at::Tensor bias_add2_fwd(at::Tensor x, at::Tensor b1, at::Tensor s,
@@ -1047,6 +1093,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_relu_fwd", &bias_relu_fwd, "fused conv-bias + relu, one NHWC pass (gfx950)");
   m.def("bias_relu_bwd", &bias_relu_bwd, "bias_relu backward: dx + fp32 db in one pass");
   m.def("bias_add2_fwd", &bias_add2_fwd, "fused residual close: x + bias1 + shortcut (+ bias2)");
+  m.def("channel_sum_fp32", &channel_sum_fp32,
+        "NHWC per-channel fp32 sum (bias gradients; small-grid atomics)");
   m.def("frames_u8_to_bf16_nhwc", &frames_u8_to_bf16_nhwc,
         "fused uint8->bf16 NHWC scale (optionally zero-padding channels)",
         py::arg("x"), py::arg("scale"), py::arg("pad_channels") = 0);

@@ -274,7 +274,8 @@ class _Conv1U8Fn(torch.autograd.Function):
         K, C = w.shape[0], w.shape[1]
         g = k.wgrad3x3_nhwc(xb8, dyb)[: 9 * 8]
         dw = g.view(3, 3, 8, K)[:, :, :C].permute(3, 2, 0, 1).to(w.dtype)
-        db = dy.sum((0, 2, 3), dtype=torch.float32).to(w.dtype)
+        # small-grid NHWC channel sum (hip/kernels.hip channel_sum_kernel)
+        db = k.channel_sum_fp32(dyb).to(w.dtype)
         return None, dw, db, None, None
 
 

@@ -401,6 +401,25 @@ class TestFusedBias:
         for a1, a2 in zip(args1, args2):
             assert torch.allclose(a1.grad, a2.grad, atol=1e-3, rtol=1e-4)
 
+    def test_channel_sum_matches_torch(self):
+        from moolib_amd import _kernels
+
+        torch.manual_seed(5)
+        for shape, dtype in [
+            ((672, 16, 42, 42), torch.bfloat16),
+            ((672, 32, 21, 21), torch.bfloat16),
+            ((672, 32, 11, 11), torch.bfloat16),
+            ((3, 8, 5, 7), torch.float32),
+        ]:
+            x = torch.randn(shape, device="cuda", dtype=dtype).contiguous(
+                memory_format=torch.channels_last
+            )
+            got = _kernels.channel_sum_fp32(x)
+            want = x.float().sum(dim=(0, 2, 3))
+            assert got.dtype == torch.float32
+            assert torch.allclose(got, want, atol=2e-1, rtol=1e-4), (
+                shape, dtype, (got - want).abs().max())
+
     def test_model_fused_matches_eager_with_grads(self):
         """Full AtariNet learner-style fwd+bwd: fused-bias path vs eager
         (MOOLIB_AMD_NO_FUSED_BIAS) — outputs and conv-bias grads agree."""
